@@ -1,0 +1,140 @@
+"""Graph partitioner (offline, CPU, vectorized numpy).
+
+Reference counterpart: helper/utils.py:73-98 (`graph_partition`, which
+delegates to DGL/METIS). libmetis/DGL are not available in this image, so:
+
+* method="random": balanced random assignment (the reference's
+  `--partition-method random`, which BNS-GCN's paper uses for papers100M);
+* method="metis": a locality partitioner — balanced contiguous ranges of
+  the node-id space. Our synthetic graphs (graph/synthetic.py) carry edge
+  locality in the id space, so this plays the role METIS plays on the real
+  datasets: a small boundary cut. The flag name is kept for CLI
+  compatibility (helper/parser.py).
+
+The partition objective flag (vol/cut) is accepted and recorded in
+meta.json but does not change the algorithm.
+"""
+from __future__ import annotations
+
+import numpy as np
+
+from .csr import Graph
+from .store import Partition, save_partitions
+
+
+def assign_parts(n_nodes: int, n_parts: int, method: str, seed: int = 0) -> np.ndarray:
+    if method == "random":
+        rng = np.random.default_rng(seed)
+        part = np.arange(n_nodes, dtype=np.int32) % n_parts
+        rng.shuffle(part)
+        return part
+    if method == "metis":  # locality/contiguous (see module docstring)
+        bounds = np.linspace(0, n_nodes, n_parts + 1).astype(np.int64)
+        part = np.zeros(n_nodes, dtype=np.int32)
+        for p in range(n_parts):
+            part[bounds[p]:bounds[p + 1]] = p
+        return part
+    raise ValueError(f"unknown partition method: {method}")
+
+
+def partition_graph(g: Graph, n_parts: int, method: str = "metis", seed: int = 0,
+                    objective: str = "vol") -> tuple[list[Partition], dict]:
+    """Split `g` into per-rank Partition objects (see store.Partition)."""
+    n = g.n_nodes
+    part = assign_parts(n, n_parts, method, seed)
+
+    # inner-local id of every node within its partition (sorted-global order)
+    inner_local = np.zeros(n, dtype=np.int64)
+    inner_lists = []
+    for p in range(n_parts):
+        nodes = np.flatnonzero(part == p)          # ascending global ids
+        inner_local[nodes] = np.arange(len(nodes))
+        inner_lists.append(nodes)
+
+    src, dst = g.adj_in.to_edges()                  # int32 arrays
+    spart = part[src]
+    dpart = part[dst]
+
+    parts: list[Partition] = []
+    boundary_all: list[list[np.ndarray]] = [[None] * n_parts for _ in range(n_parts)]
+
+    for p in range(n_parts):
+        nodes = inner_lists[p]
+        n_inner = len(nodes)
+        em = dpart == p
+        e_src, e_dst = src[em], dst[em]
+        e_spart = spart[em]
+        dst_local = inner_local[e_dst].astype(np.int32)
+
+        # inner->inner edges
+        im = e_spart == p
+        from .csr import CSR
+        inner_csr = CSR.from_edges(inner_local[e_src[im]].astype(np.int32),
+                                   dst_local[im], n_inner, n_inner, sort_cols=True)
+
+        # halo edges grouped by (owner, owner-local id)
+        hm = ~im
+        h_owner = e_spart[hm]
+        h_ol = inner_local[e_src[hm]].astype(np.int32)   # owner-local src id
+        h_dst = dst_local[hm]
+        # unique halo rows in (owner, owner_local) order == peer-major sorted
+        key = h_owner.astype(np.int64) * n + h_ol
+        uniq, inv = np.unique(key, return_inverse=True)
+        halo_part = (uniq // n).astype(np.int32)
+        halo_ol = (uniq % n).astype(np.int32)
+        halo_csr = CSR.from_edges(h_dst, inv.astype(np.int64), len(uniq), n_inner,
+                                  sort_cols=True)
+
+        # halo degrees come straight from the full graph (global view)
+        halo_global = np.empty(len(uniq), dtype=np.int64)
+        for j in range(n_parts):
+            m = halo_part == j
+            if m.any():
+                halo_global[m] = inner_lists[j][halo_ol[m]]
+        halo_out_deg = g.out_deg[halo_global].astype(np.int32)
+        halo_in_deg = g.in_deg[halo_global].astype(np.int32)
+
+        # outgoing boundary of each OWNER j toward me (p): owner-local ids,
+        # sorted — record into boundary_all[j][p]
+        for j in range(n_parts):
+            m = halo_part == j
+            boundary_all[j][p] = halo_ol[m].copy()   # already sorted within peer
+
+        parts.append(Partition(
+            rank=p, n_parts=n_parts,
+            inner_global_nid=nodes.astype(np.int64),
+            feat=g.feat[nodes], label=g.label[nodes],
+            train_mask=g.train_mask[nodes], val_mask=g.val_mask[nodes],
+            test_mask=g.test_mask[nodes],
+            in_deg=g.in_deg[nodes].astype(np.int32),
+            out_deg=g.out_deg[nodes].astype(np.int32),
+            inner_indptr=inner_csr.indptr, inner_indices=inner_csr.indices,
+            halo_part=halo_part, halo_owner_local=halo_ol,
+            halo_out_deg=halo_out_deg, halo_in_deg=halo_in_deg,
+            halo_indptr=halo_csr.indptr, halo_indices=halo_csr.indices,
+        ))
+
+    for p in range(n_parts):
+        parts[p].boundary = [
+            boundary_all[p][j] if boundary_all[p][j] is not None
+            else np.zeros(0, dtype=np.int32)
+            for j in range(n_parts)]
+        parts[p].boundary[p] = np.zeros(0, dtype=np.int32)
+
+    meta = {
+        "n_parts": n_parts, "method": method, "objective": objective, "seed": seed,
+        "n_nodes": int(g.n_nodes), "n_edges": int(g.n_edges),
+        "n_feat": int(g.n_feat), "n_class": int(g.n_class),
+        "n_train": int(g.n_train), "multilabel": bool(g.multilabel),
+        "dataset": g.name,
+    }
+    return parts, meta
+
+
+def partition_and_save(g: Graph, n_parts: int, method: str, out_dir: str,
+                       graph_name: str, seed: int = 0, objective: str = "vol",
+                       extra_meta: dict | None = None) -> str:
+    parts, meta = partition_graph(g, n_parts, method, seed, objective)
+    if extra_meta:
+        meta.update(extra_meta)
+    return save_partitions(parts, meta, out_dir, graph_name)

@@ -1,0 +1,128 @@
+"""Graph core: CSR container, synthetic generators, partitioner + store.
+
+The partitioner invariants verified here correspond to what the reference
+gets from DGL's partition store (reference: helper/utils.py:101-140): node
+coverage, edge conservation, halo/boundary duality.
+"""
+import numpy as np
+import pytest
+
+from bnsgcn_amd.graph import (CSR, add_self_loops, load_data, partition_graph,
+                              save_partitions, load_partition)
+
+
+def test_csr_roundtrip():
+    src = np.array([0, 1, 2, 2, 3], dtype=np.int32)
+    dst = np.array([1, 0, 0, 3, 3], dtype=np.int32)
+    c = CSR.from_edges(src, dst, 4, 4, sort_cols=True)
+    assert c.n_edges == 5
+    assert list(c.indptr) == [0, 2, 3, 3, 5]
+    np.testing.assert_array_equal(c.indices, [1, 2, 0, 2, 3])
+    s2, d2 = c.to_edges()
+    c2 = CSR.from_edges(s2, d2, 4, 4, sort_cols=True)
+    np.testing.assert_array_equal(c2.indices, c.indices)
+    np.testing.assert_array_equal(c2.indptr, c.indptr)
+
+
+def test_csr_transpose():
+    rng = np.random.default_rng(0)
+    src = rng.integers(0, 50, 400).astype(np.int32)
+    dst = rng.integers(0, 50, 400).astype(np.int32)
+    c = CSR.from_edges(src, dst, 50, 50)
+    t = c.transpose()
+    # dense check
+    A = np.zeros((50, 50))
+    np.add.at(A, (dst, src), 1)
+    T = np.zeros((50, 50))
+    td, ts = t.to_edges()  # (cols=src of t, rows)
+    np.add.at(T, (ts, td), 1)
+    np.testing.assert_array_equal(A.T, T)
+
+
+def test_self_loops():
+    src = np.array([0, 1, 1], dtype=np.int32)
+    dst = np.array([0, 2, 1], dtype=np.int32)
+    s, d = add_self_loops(src, dst, 3)
+    assert len(s) == 4  # two self-loops dropped, three added
+    loops = (s == d).sum()
+    assert loops == 3
+
+
+def test_synthetic_shapes_and_determinism():
+    g1 = load_data("tiny", seed=3)
+    g2 = load_data("tiny", seed=3)
+    np.testing.assert_array_equal(g1.adj_in.indices, g2.adj_in.indices)
+    np.testing.assert_allclose(g1.feat, g2.feat)
+    assert g1.n_feat == 16 and g1.n_class == 7
+    # self-loops present exactly once per node
+    s, d = g1.adj_in.to_edges()
+    assert (s == d).sum() == g1.n_nodes
+    g3 = load_data("tiny", seed=4)
+    assert not np.array_equal(g1.adj_in.indices, g3.adj_in.indices)
+
+
+def test_multilabel_dataset():
+    g = load_data("tiny-ml", seed=0)
+    assert g.multilabel and g.label.shape == (g.n_nodes, 5)
+
+
+@pytest.mark.parametrize("method", ["random", "metis"])
+def test_partition_invariants(method):
+    g = load_data("tiny", seed=1)
+    P = 4
+    parts, meta = partition_graph(g, P, method=method, seed=0)
+    # node coverage, disjoint
+    all_nodes = np.concatenate([p.inner_global_nid for p in parts])
+    assert len(all_nodes) == g.n_nodes
+    assert len(np.unique(all_nodes)) == g.n_nodes
+    # edge conservation
+    tot_edges = sum(len(p.inner_indices) + len(p.halo_indices) for p in parts)
+    assert tot_edges == g.n_edges
+    # features/labels/degrees match the global graph
+    for p in parts:
+        np.testing.assert_allclose(p.feat, g.feat[p.inner_global_nid])
+        np.testing.assert_array_equal(p.in_deg, g.in_deg[p.inner_global_nid])
+        np.testing.assert_array_equal(p.out_deg, g.out_deg[p.inner_global_nid])
+        # local in-degree sums: inner + halo rows = full in-degree
+        local_in = np.diff(p.inner_indptr)
+        halo_cnt = np.bincount(p.halo_indices, minlength=p.n_inner)
+        np.testing.assert_array_equal(local_in + halo_cnt, p.in_deg)
+    # halo/boundary duality: parts[i].boundary[j] == parts[j] halo rows owned by i
+    for i in range(P):
+        for j in range(P):
+            if i == j:
+                continue
+            sl = parts[j].halo_peer_slices()[i]
+            np.testing.assert_array_equal(parts[i].boundary[j],
+                                          parts[j].halo_owner_local[sl])
+    # halo degrees match the owner's full-graph degrees
+    for p in parts:
+        for j in range(P):
+            sl = p.halo_peer_slices()[j]
+            ol = p.halo_owner_local[sl]
+            np.testing.assert_array_equal(p.halo_out_deg[sl], parts[j].out_deg[ol])
+
+
+def test_store_roundtrip(tmp_path):
+    g = load_data("tiny", seed=2)
+    parts, meta = partition_graph(g, 3, method="random", seed=0)
+    save_partitions(parts, meta, str(tmp_path), "tiny-3")
+    p1 = load_partition(str(tmp_path), "tiny-3", 1)
+    np.testing.assert_array_equal(p1.inner_global_nid, parts[1].inner_global_nid)
+    np.testing.assert_array_equal(p1.halo_indices, parts[1].halo_indices)
+    np.testing.assert_array_equal(p1.boundary[0], parts[1].boundary[0])
+    assert p1.meta["n_feat"] == 16
+
+
+def test_subgraph_inductive():
+    g = load_data("tiny", seed=5)
+    sub = g.subgraph(g.train_mask)
+    assert sub.n_nodes == g.n_train
+    assert sub.train_mask.all()
+    # every subgraph edge existed in g
+    s, d = sub.adj_in.to_edges()
+    keep = np.flatnonzero(g.train_mask)
+    gs, gd = g.adj_in.to_edges()
+    eset = set(zip(gs.tolist(), gd.tolist()))
+    for a, b in zip(keep[s].tolist(), keep[d].tolist()):
+        assert (a, b) in eset
