@@ -1,0 +1,206 @@
+"""Autograd-wired compute ops with CPU(torch-reference) / GPU(HIP) dispatch.
+
+Each op dispatches per-tensor-device: CPU → ops/reference.py; CUDA(HIP) →
+the gfx950 extension (required — no silent eager fallback on GPU, see
+ops/_ext.py). Backward passes reuse the same primitives on precomputed
+transposed CSRs, so one kernel family serves both directions
+(reference counterparts: SURVEY.md §2.3 K1-K6, K13-K14).
+"""
+from __future__ import annotations
+
+import torch
+from torch.autograd import Function
+
+from . import reference as ref
+from ._ext import get_ext, use_hip
+
+
+# ---------------------------------------------------------------- raw ops
+
+def spmm_sum_raw(indptr, indices, x, src_scale=None, dst_scale=None, out=None):
+    if use_hip(x):
+        return get_ext().spmm_sum(indptr, indices, x, src_scale, dst_scale, out)
+    return ref.spmm_sum(indptr, indices, x, src_scale, dst_scale, out)
+
+
+def spmm_edge_raw(indptr, indices, eweight, x, out=None):
+    if use_hip(x):
+        return get_ext().spmm_edge_sum(indptr, indices, eweight, x, out)
+    return ref.spmm_edge_sum(indptr, indices, eweight, x, out)
+
+
+def sddmm_dot_raw(indptr, indices, a_dst, b_src):
+    if use_hip(a_dst):
+        return get_ext().sddmm_dot(indptr, indices, a_dst, b_src)
+    return ref.sddmm_dot(indptr, indices, a_dst, b_src)
+
+
+def sddmm_add_raw(indptr, indices, el_src, er_dst):
+    if use_hip(el_src):
+        return get_ext().sddmm_add(indptr, indices, el_src, er_dst)
+    return ref.sddmm_add(indptr, indices, el_src, er_dst)
+
+
+def segment_softmax_raw(indptr, logits):
+    if use_hip(logits):
+        return get_ext().segment_softmax(indptr, logits)
+    return ref.segment_softmax(indptr, logits)
+
+
+def segment_softmax_bwd_raw(indptr, alpha, grad_alpha):
+    if use_hip(alpha):
+        return get_ext().segment_softmax_backward(indptr, alpha, grad_alpha)
+    return ref.segment_softmax_backward(indptr, alpha, grad_alpha)
+
+
+def pack_rows_raw(x, idx, scale=None):
+    if use_hip(x):
+        return get_ext().pack_rows(x, idx, scale)
+    return ref.pack_rows(x, idx, scale)
+
+
+def scatter_add_rows_raw(out, idx, src, scale=None):
+    if use_hip(out):
+        get_ext().scatter_add_rows(out, idx, src, scale)
+        return out
+    return ref.scatter_add_rows(out, idx, src, scale)
+
+
+# ----------------------------------------------------------- autograd ops
+
+class _SpMMSum(Function):
+    """y[r] = dst_scale[r] * Σ_{c∈row r} src_scale[c] * x[c]  (K1/K2 + K18).
+
+    Backward runs the SAME kernel on the precomputed transposed CSR with the
+    scale roles swapped — no atomics anywhere (SURVEY.md §7 'per-epoch block
+    assembly' design)."""
+
+    @staticmethod
+    def forward(ctx, x, indptr, indices, indptr_t, indices_t, src_scale, dst_scale):
+        ctx.save_for_backward(indptr_t, indices_t, src_scale, dst_scale)
+        return spmm_sum_raw(indptr, indices, x, src_scale, dst_scale)
+
+    @staticmethod
+    def backward(ctx, grad):
+        indptr_t, indices_t, src_scale, dst_scale = ctx.saved_tensors
+        gx = spmm_sum_raw(indptr_t, indices_t, grad.contiguous(),
+                          src_scale=dst_scale, dst_scale=src_scale)
+        return gx, None, None, None, None, None, None
+
+
+def spmm_sum(x, indptr, indices, indptr_t, indices_t,
+             src_scale=None, dst_scale=None):
+    return _SpMMSum.apply(x, indptr, indices, indptr_t, indices_t,
+                          src_scale, dst_scale)
+
+
+class _SpMMEdge(Function):
+    """y[r,h] = Σ_{e∈row r} w[e,h] * x[col_e,h,:]  (GAT aggregate, K4/K5)."""
+
+    @staticmethod
+    def forward(ctx, x, w, indptr, indices, indptr_t, indices_t, eperm_t):
+        ctx.save_for_backward(x, w, indptr, indices, indptr_t, indices_t, eperm_t)
+        return spmm_edge_raw(indptr, indices, w, x)
+
+    @staticmethod
+    def backward(ctx, grad):
+        x, w, indptr, indices, indptr_t, indices_t, eperm_t = ctx.saved_tensors
+        grad = grad.contiguous()
+        gx = gw = None
+        if ctx.needs_input_grad[0]:
+            gx = spmm_edge_raw(indptr_t, indices_t, w[eperm_t], grad)
+        if ctx.needs_input_grad[1]:
+            gw = sddmm_dot_raw(indptr, indices, grad, x)
+        return gx, gw, None, None, None, None, None
+
+
+def spmm_edge_sum(x, w, indptr, indices, indptr_t, indices_t, eperm_t):
+    return _SpMMEdge.apply(x, w, indptr, indices, indptr_t, indices_t, eperm_t)
+
+
+class _SDDMMAdd(Function):
+    """logits[e,h] = el[col_e,h] + er[row_e,h]  (u_add_v SDDMM)."""
+
+    @staticmethod
+    def forward(ctx, el, er, indptr, indices, indptr_t, indices_t, eperm_t):
+        ctx.save_for_backward(indptr, indices, indptr_t, indices_t, eperm_t)
+        ctx.n_src, ctx.n_dst = el.shape[0], er.shape[0]
+        return sddmm_add_raw(indptr, indices, el, er)
+
+    @staticmethod
+    def backward(ctx, grad):
+        indptr, indices, indptr_t, indices_t, eperm_t = ctx.saved_tensors
+        grad = grad.contiguous()
+        g_el = g_er = None
+        ones = None
+        if ctx.needs_input_grad[0]:
+            # d el[c] = Σ_{e: col_e=c} grad[e]: edge-weighted spmm on the
+            # transpose with x = ones → equivalently segment-sum by col.
+            g_el = torch.zeros(ctx.n_src, grad.shape[1], dtype=grad.dtype,
+                               device=grad.device)
+            g_el.index_add_(0, indices.long(), grad)
+        if ctx.needs_input_grad[1]:
+            g_er = torch.zeros(ctx.n_dst, grad.shape[1], dtype=grad.dtype,
+                               device=grad.device)
+            row = torch.repeat_interleave(
+                torch.arange(ctx.n_dst, device=grad.device),
+                indptr[1:] - indptr[:-1])
+            g_er.index_add_(0, row, grad)
+        return g_el, g_er, None, None, None, None, None
+
+
+def sddmm_add(el, er, indptr, indices, indptr_t, indices_t, eperm_t):
+    return _SDDMMAdd.apply(el, er, indptr, indices, indptr_t, indices_t, eperm_t)
+
+
+class _SegmentSoftmax(Function):
+    @staticmethod
+    def forward(ctx, logits, indptr):
+        alpha = segment_softmax_raw(indptr, logits)
+        ctx.save_for_backward(alpha, indptr)
+        return alpha
+
+    @staticmethod
+    def backward(ctx, grad):
+        alpha, indptr = ctx.saved_tensors
+        return segment_softmax_bwd_raw(indptr, alpha, grad.contiguous()), None
+
+
+def segment_softmax(logits, indptr):
+    return _SegmentSoftmax.apply(logits, indptr)
+
+
+class _Linear(Function):
+    """XW^T + b on the hand-written fp32 MFMA GEMM (K6) when on GPU."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        if use_hip(x):
+            return get_ext().gemm_nt_bias(x, weight, bias)
+        return torch.nn.functional.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, grad):
+        x, weight = ctx.saved_tensors
+        grad = grad.contiguous()
+        gx = gw = gb = None
+        if use_hip(x):
+            e = get_ext()
+            if ctx.needs_input_grad[0]:
+                gx = e.gemm_nn(grad, weight)
+            if ctx.needs_input_grad[1]:
+                gw = e.gemm_tn(grad, x)
+        else:
+            if ctx.needs_input_grad[0]:
+                gx = grad @ weight
+            if ctx.needs_input_grad[1]:
+                gw = grad.t() @ x
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = grad.sum(0)
+        return gx, gw, gb
+
+
+def linear(x, weight, bias=None):
+    return _Linear.apply(x, weight, bias)
