@@ -1,0 +1,142 @@
+"""GraphContext — what a model forward needs to know about the graph.
+
+Two flavors:
+* partition context (training): inner CSRs + HaloPlan (+ a full-halo state
+  for use_pp precompute / GAT layer 0);
+* full-graph context (evaluation / single-process): plain CSRs, no plan.
+
+This replaces the reference's per-epoch DGL heterograph argument
+(train.py:392-404, module/layer.py graph arg) with precomputed device
+tensors; the per-epoch part lives in HaloPlan.set_epoch.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from ..graph.store import Partition
+from ..ops.csr_torch import transpose_csr, merge_csr
+from ..ops.functional import spmm_sum_raw
+from ..parallel.halo import partition_aggregate
+from ..parallel.plan import EpochState, HaloPlan
+
+
+def _inv_sqrt(x: torch.Tensor) -> torch.Tensor:
+    return 1.0 / torch.sqrt(x.clamp_min(1.0))
+
+
+def _inv(x: torch.Tensor) -> torch.Tensor:
+    return 1.0 / x.clamp_min(1.0)
+
+
+class GraphContext:
+    def __init__(self, indptr, indices, in_deg, out_deg, device,
+                 plan: HaloPlan | None = None):
+        dev = torch.device(device)
+        self.plan = plan
+        self.indptr = indptr.to(dev)
+        self.indices = indices.to(dev)
+        tip, tix, _ = transpose_csr(self.indptr, self.indices,
+                                    n_cols=self.indptr.numel() - 1
+                                    if plan is None else plan.n_inner)
+        self.t_indptr, self.t_indices = tip, tix
+        in_deg = in_deg.to(dev).float()
+        out_deg = out_deg.to(dev).float()
+        self.in_norm_inv = _inv_sqrt(in_deg)       # GCN dst scale
+        self.out_norm_inv = _inv_sqrt(out_deg)     # GCN src scale
+        self.in_deg_inv = _inv(in_deg)             # SAGE mean dst scale
+        self.n_rows = self.indptr.numel() - 1
+        # lazily built per-epoch GAT block (combined inner+halo CSR)
+        self._gat_cache: tuple[int, tuple] | None = None
+        # full-halo (p=1.0) exchange state for precompute / GAT layer 0
+        self._full_state: EpochState | None = None
+
+    # ---------------------------------------------------------------- train
+    @classmethod
+    def for_partition(cls, part: Partition, plan: HaloPlan, device) -> "GraphContext":
+        return cls(torch.from_numpy(part.inner_indptr),
+                   torch.from_numpy(part.inner_indices),
+                   torch.from_numpy(part.in_deg),
+                   torch.from_numpy(part.out_deg), device, plan=plan)
+
+    # ----------------------------------------------------------------- eval
+    @classmethod
+    def for_full_graph(cls, csr_indptr, csr_indices, in_deg, out_deg, device
+                       ) -> "GraphContext":
+        return cls(csr_indptr, csr_indices, in_deg, out_deg, device, plan=None)
+
+    @property
+    def inner_csrs(self):
+        return (self.indptr, self.indices, self.t_indptr, self.t_indices)
+
+    def aggregate(self, x: torch.Tensor, mode: str) -> torch.Tensor:
+        """mode='gcn': symmetric-normalized sum (reference layer.py:32-38);
+        mode='mean': in-degree mean with FULL-graph degrees
+        (reference layer.py:85-92 — degrees precomputed before
+        partitioning, utils.py:92-93, so the sampled estimator stays
+        unbiased after the 1/ratio pack scale)."""
+        if mode == "gcn":
+            src, dst, halo_src = self.out_norm_inv, self.in_norm_inv, True
+        elif mode == "mean":
+            src, dst, halo_src = None, self.in_deg_inv, False
+        else:
+            raise ValueError(mode)
+        if self.plan is None:
+            from ..ops.functional import spmm_sum
+            return spmm_sum(x, *self.inner_csrs, src_scale=src, dst_scale=dst)
+        return partition_aggregate(x, self.plan, self.inner_csrs, src, dst, halo_src)
+
+    # ------------------------------------------------------------- GAT block
+    def gat_block(self):
+        """Combined (inner ∪ sampled-halo) dst-indexed CSR with source ids in
+        [0, n_inner + R): cols < n_inner are inner sources, cols >= n_inner
+        index the packed received rows. Rebuilt per epoch, cached."""
+        if self.plan is None:
+            ip, ix = self.indptr, self.indices
+            tip, tix, eperm = transpose_csr(ip, ix, self.n_rows)
+            return ip, ix, tip, tix, eperm, 0
+        st = self.plan.state
+        if self._gat_cache is not None and self._gat_cache[0] == st.epoch:
+            return self._gat_cache[1]
+        n_inner = self.plan.n_inner
+        ip, ix = merge_csr(self.indptr, self.indices,
+                           st.halo_fwd_indptr, st.halo_fwd_indices,
+                           col_offset2=n_inner)
+        R = sum(st.recv_counts)
+        tip, tix, eperm = transpose_csr(ip, ix, n_inner + R)
+        block = (ip, ix, tip, tix, eperm, R)
+        self._gat_cache = (st.epoch, block)
+        return block
+
+    def gat_block_full(self):
+        """Combined (inner ∪ FULL halo) block for GAT layer 0 under use_pp:
+        every halo row participates, no sampling (reference model.py:118-120
+        uses the full merged precompute tensor)."""
+        if getattr(self, "_gat_full_cache", None) is not None:
+            return self._gat_full_cache
+        n_inner = self.plan.n_inner
+        fip, fix, _ = transpose_csr(self.plan.halo_indptr,
+                                    self.plan.halo_indices, n_inner)
+        ip, ix = merge_csr(self.indptr, self.indices, fip, fix,
+                           col_offset2=n_inner)
+        n_halo = self.plan.halo_indptr.numel() - 1
+        tip, tix, eperm = transpose_csr(ip, ix, n_inner + n_halo)
+        self._gat_full_cache = (ip, ix, tip, tix, eperm, n_halo)
+        return self._gat_full_cache
+
+    # -------------------------------------------------- full-halo exchange
+    def full_state(self) -> EpochState:
+        """Exchange plan at sampling rate 1.0 (use_pp precompute, reference
+        train.py:170-211, and GAT layer 0 under use_pp)."""
+        assert self.plan is not None
+        if self._full_state is None:
+            full = HaloPlan.__new__(HaloPlan)
+            full.__dict__.update(self.plan.__dict__)
+            full.rate = 1.0
+            full.unit_ratio = True
+            full.send_size = list(full.n_out)
+            full.recv_size = list(full.n_in)
+            full._state = None
+            full._static = True
+            self._full_state = full.set_epoch(0)
+        return self._full_state

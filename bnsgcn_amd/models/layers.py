@@ -1,0 +1,132 @@
+"""GNN layers over a GraphContext.
+
+Math matches the reference exactly (module/layer.py for GCN/SAGE; DGL
+GATConv as instantiated at module/model.py:102 for GAT), but expressed on
+our partition-aggregate / SDDMM / segment-softmax primitives:
+
+* GCNLayer   — h' = Linear( in_norm^-1 · Σ_src out_norm^-1 · h )
+               (layer.py:32-38; use_pp fast path layer.py:29-30)
+* SAGELayer  — h' = Linear1(h_dst) + Linear2( (Σ_src h)/in_deg )
+               (layer.py:85-92; use_pp path = one Linear over [h ‖ agg],
+               layer.py:59,82-83)
+* GATLayer   — multi-head attention: z=fc(h); e=LeakyReLU(a_l·z_src+a_r·z_dst);
+               α=edge-softmax_dst(e); out=Σ α·z_src + bias (DGL GATConv
+               semantics with feat_drop/attn_drop, negative_slope=0.2).
+
+Parameter init mirrors the reference (uniform ±1/sqrt(fan_in),
+layer.py:20-24,65-77; GATConv uses xavier with gain=sqrt(2)).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+from torch import nn
+
+from ..ops import functional as F
+from .context import GraphContext
+
+
+def _uniform_init(*tensors, fan_in):
+    stdv = 1.0 / math.sqrt(fan_in)
+    for t in tensors:
+        nn.init.uniform_(t, -stdv, stdv)
+
+
+class GCNLayer(nn.Module):
+    def __init__(self, in_feats, out_feats, bias=True, use_pp=False):
+        super().__init__()
+        self.use_pp = use_pp
+        self.weight = nn.Parameter(torch.empty(out_feats, in_feats))
+        self.bias = nn.Parameter(torch.empty(out_feats)) if bias else None
+        _uniform_init(self.weight, fan_in=in_feats)
+        if self.bias is not None:
+            _uniform_init(self.bias, fan_in=in_feats)
+
+    def forward(self, ctx: GraphContext, x):
+        if self.training and self.use_pp:
+            return F.linear(x, self.weight, self.bias)
+        h = ctx.aggregate(x, "gcn")
+        return F.linear(h, self.weight, self.bias)
+
+
+class SAGELayer(nn.Module):
+    def __init__(self, in_feats, out_feats, bias=True, use_pp=False):
+        super().__init__()
+        self.use_pp = use_pp
+        if use_pp:
+            self.weight = nn.Parameter(torch.empty(out_feats, 2 * in_feats))
+            self.bias = nn.Parameter(torch.empty(out_feats)) if bias else None
+            _uniform_init(self.weight, fan_in=2 * in_feats)
+            if self.bias is not None:
+                _uniform_init(self.bias, fan_in=2 * in_feats)
+        else:
+            self.weight1 = nn.Parameter(torch.empty(out_feats, in_feats))
+            self.weight2 = nn.Parameter(torch.empty(out_feats, in_feats))
+            self.bias1 = nn.Parameter(torch.empty(out_feats)) if bias else None
+            self.bias2 = nn.Parameter(torch.empty(out_feats)) if bias else None
+            _uniform_init(self.weight1, self.weight2, fan_in=in_feats)
+            if bias:
+                _uniform_init(self.bias1, self.bias2, fan_in=in_feats)
+
+    def forward(self, ctx: GraphContext, x):
+        if self.training and self.use_pp:
+            # x = [feat ‖ precomputed neighbor mean], width 2F
+            return F.linear(x, self.weight, self.bias)
+        ah = ctx.aggregate(x, "mean")
+        if self.use_pp:  # eval path of a pp layer (reference layer.py:98-100)
+            return F.linear(torch.cat((x, ah), dim=1), self.weight, self.bias)
+        return (F.linear(x, self.weight1, self.bias1)
+                + F.linear(ah, self.weight2, self.bias2))
+
+
+class GATLayer(nn.Module):
+    def __init__(self, in_feats, out_feats, heads, feat_drop=0.0, attn_drop=0.0,
+                 negative_slope=0.2, bias=True, use_pp=False):
+        super().__init__()
+        self.heads, self.out_feats = heads, out_feats
+        self.use_pp = use_pp
+        self.fc = nn.Parameter(torch.empty(heads * out_feats, in_feats))
+        self.attn_l = nn.Parameter(torch.empty(1, heads, out_feats))
+        self.attn_r = nn.Parameter(torch.empty(1, heads, out_feats))
+        self.bias = nn.Parameter(torch.zeros(heads * out_feats)) if bias else None
+        self.feat_drop = nn.Dropout(feat_drop)
+        self.attn_drop = nn.Dropout(attn_drop)
+        self.negative_slope = negative_slope
+        gain = math.sqrt(2.0)
+        nn.init.xavier_normal_(self.fc, gain=gain)
+        nn.init.xavier_normal_(self.attn_l, gain=gain)
+        nn.init.xavier_normal_(self.attn_r, gain=gain)
+
+    def forward(self, ctx: GraphContext, x, halo_feat: torch.Tensor | None = None):
+        """x: [n_local, F]. In partition mode, halo sources are fetched via
+        halo_exchange — except when `halo_feat` is given (GAT layer 0 under
+        use_pp: the FULL unsampled halo features captured at precompute,
+        reference model.py:118-120 / train.py:208-209)."""
+        from ..parallel.halo import halo_exchange
+
+        h = self.feat_drop(x)
+        H, D = self.heads, self.out_feats
+        if ctx.plan is not None:
+            if halo_feat is not None:
+                src_extra = self.feat_drop(halo_feat)
+                ip, ix, tip, tix, eperm, _ = ctx.gat_block_full()
+            else:
+                src_extra = halo_exchange(h, ctx.plan)
+                ip, ix, tip, tix, eperm, _ = ctx.gat_block()
+            src = torch.cat((h, src_extra), dim=0)
+        else:
+            src = h
+            ip, ix, tip, tix, eperm, _ = ctx.gat_block()
+        z_src = F.linear(src, self.fc).view(-1, H, D)
+        z_dst = z_src[:x.shape[0]]
+        el = (z_src * self.attn_l).sum(-1)          # [n_src, H]
+        er = (z_dst * self.attn_r).sum(-1)          # [n_dst, H]
+        logits = F.sddmm_add(el, er, ip, ix, tip, tix, eperm)
+        logits = torch.nn.functional.leaky_relu(logits, self.negative_slope)
+        alpha = F.segment_softmax(logits, ip)
+        alpha = self.attn_drop(alpha)
+        out = F.spmm_edge_sum(z_src, alpha, ip, ix, tip, tix, eperm)
+        if self.bias is not None:
+            out = out + self.bias.view(1, H, D)
+        return out

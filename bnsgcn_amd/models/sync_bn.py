@@ -1,0 +1,69 @@
+"""Distributed SyncBatchNorm (reference: module/sync_bn.py).
+
+Forward all-reduces per-feature Σx and Σx² and normalizes with the GLOBAL
+mean/var over `whole_size` rows (= global train-node count — correct when
+partition row counts sum to whole_size, i.e. inductive mode; SURVEY.md
+§2.5.8). Backward all-reduces dbias/dweight and forms dx analytically.
+Selected by --norm batch.
+"""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+from torch import nn
+from torch.autograd import Function
+
+
+def _maybe_all_reduce(t: torch.Tensor) -> torch.Tensor:
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t
+
+
+class _SyncBNFunc(Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, whole_size, running_mean, running_var,
+                training, momentum, eps):
+        if not training:
+            mean, var = running_mean, running_var
+        else:
+            stats = torch.stack((x.sum(0), (x * x).sum(0)))
+            _maybe_all_reduce(stats)
+            mean = stats[0] / whole_size
+            var = (stats[1] - mean * stats[0]) / whole_size
+            with torch.no_grad():
+                running_mean.mul_(1 - momentum).add_(mean * momentum)
+                running_var.mul_(1 - momentum).add_(var * momentum)
+        std = torch.sqrt(var + eps)
+        x_hat = (x - mean) / std
+        if training:
+            ctx.save_for_backward(x_hat, weight, std)
+            ctx.whole_size = whole_size
+        return x_hat * weight + bias
+
+    @staticmethod
+    def backward(ctx, grad):
+        x_hat, weight, std = ctx.saved_tensors
+        n = ctx.whole_size
+        red = torch.stack((grad.sum(0), (grad * x_hat).sum(0)))
+        _maybe_all_reduce(red)
+        dbias, dweight = red[0], red[1]
+        dx = (weight / n) / std * (n * grad - dbias - x_hat * dweight)
+        return dx, dweight, dbias, None, None, None, None, None, None
+
+
+class SyncBatchNorm(nn.Module):
+    def __init__(self, num_features, whole_size, eps=1e-5, momentum=0.1):
+        super().__init__()
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.whole_size = whole_size
+        self.eps = eps
+        self.momentum = momentum
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+
+    def forward(self, x):
+        return _SyncBNFunc.apply(x, self.weight, self.bias, self.whole_size,
+                                 self.running_mean, self.running_var,
+                                 self.training, self.momentum, self.eps)

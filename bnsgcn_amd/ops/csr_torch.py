@@ -31,6 +31,33 @@ def transpose_csr(indptr: torch.Tensor, indices: torch.Tensor, n_cols: int
     return indptr_t, indices_t, eperm
 
 
+def merge_csr(ip1: torch.Tensor, ix1: torch.Tensor,
+              ip2: torch.Tensor, ix2: torch.Tensor,
+              col_offset2: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Row-wise union of two CSRs over the same row set; the second CSR's
+    columns are shifted by `col_offset2` (used to append halo sources after
+    inner sources for the GAT attention block). Within each output row, the
+    first CSR's edges precede the second's."""
+    device = ix1.device
+    n = ip1.numel() - 1
+    l1 = row_lengths(ip1)
+    l2 = row_lengths(ip2)
+    ip = torch.zeros(n + 1, dtype=ip1.dtype, device=device)
+    torch.cumsum(l1 + l2, 0, out=ip[1:])
+    ix = torch.empty(int(ip[-1]), dtype=ix1.dtype, device=device)
+    rows = torch.arange(n, device=device)
+    # positions of csr1 edges: ip[r] + offset-within-row
+    if ix1.numel():
+        r1 = torch.repeat_interleave(rows, l1)
+        e1 = torch.arange(ix1.numel(), device=device)
+        ix[ip[r1] + (e1 - ip1[r1])] = ix1
+    if ix2.numel():
+        r2 = torch.repeat_interleave(rows, l2)
+        e2 = torch.arange(ix2.numel(), device=device)
+        ix[ip[r2] + l1[r2] + (e2 - ip2[r2])] = (ix2 + col_offset2).to(ix1.dtype)
+    return ip, ix
+
+
 def gather_rows_csr(indptr: torch.Tensor, indices: torch.Tensor,
                     rows: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     """Sub-CSR containing the listed rows, re-numbered 0..len(rows)-1 in the

@@ -1,0 +1,315 @@
+"""Training runtime: setup, precompute, epoch loop, eval, checkpoint.
+
+Reference counterpart: train.py (run/init_processes and its 15 helpers —
+call stacks in SURVEY.md §3). Key structural differences (MI355X-first):
+
+* no per-epoch DGL graph rebuild — HaloPlan.set_epoch does CSR gathers;
+* no NODE-id exchange — shared Philox sampling (parallel/plan.py);
+* the halo all-to-all overlaps inner-edge SpMM (parallel/halo.py);
+* gradient sync is one bucketed RCCL all-reduce (parallel/reducer.py).
+Observable behavior kept: loss = sum-reduced CE/BCE over local train rows
+scaled 1/n_train at grad sync (train.py:358-361, reducer.py:34), per-epoch
+log line, checkpoint names `checkpoint/<graph>_p<rate>_<epoch>.pth.tar` /
+`<graph>_final.pth.tar` (train.py:428,452), full-graph eval on rank 0 in a
+background thread (train.py:434-442).
+"""
+from __future__ import annotations
+
+import copy
+import os
+import time
+from concurrent.futures import ThreadPoolExecutor
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..graph import load_data, load_partition, partition_and_save
+from ..graph.store import Partition
+from ..models.context import GraphContext
+from ..models.models import create_model
+from ..ops.functional import pack_rows_raw, spmm_sum_raw
+from ..parallel import GradReducer, HaloPlan, all_to_all_rows, init_distributed
+from ..utils.timer import comm_timer
+from .config import graph_name_of
+
+
+# --------------------------------------------------------------- offline
+
+def prepare_partitions(args) -> str:
+    """Rank-0 offline step (reference main.py:26-31 + graph_partition)."""
+    name = graph_name_of(args)
+    d = os.path.join(args.partition_dir, name)
+    if args.skip_partition and os.path.exists(os.path.join(d, "meta.json")):
+        return d
+    g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
+    if args.inductive:
+        g = g.subgraph(g.train_mask, name=g.name)
+    extra = {"inductive": args.inductive, "dataset_seed": args.seed,
+             "data_scale": args.data_scale,
+             "full_n_nodes": g.n_nodes, "full_n_edges": g.n_edges}
+    partition_and_save(g, args.n_partitions, args.partition_method,
+                       args.partition_dir, name, seed=args.seed,
+                       objective=args.partition_obj, extra_meta=extra)
+    return d
+
+
+# ------------------------------------------------------------- rank setup
+
+class RankState:
+    """Everything one rank needs for the epoch loop."""
+
+    def __init__(self, part: Partition, args, device):
+        self.device = torch.device(device)
+        self.part = part
+        self.args = args
+        dev = self.device
+        self.feat = torch.from_numpy(part.feat).to(dev)
+        lab = torch.from_numpy(part.label)
+        self.label = lab.to(dev)
+        self.train_mask = torch.from_numpy(part.train_mask).to(dev)
+        self.plan = HaloPlan(part, args.sampling_rate, seed=args.seed, device=dev,
+                             unit_ratio=(args.model == "gat"
+                                         and not args.gat_ratio_scale))
+        self.ctx = GraphContext.for_partition(part, self.plan, dev)
+        self.halo_feat0 = None     # GAT use_pp layer-0 full halo features
+        self.n_train_global = int(part.meta["n_train"])
+        self.part_train = int(part.train_mask.sum())
+
+    # ---------------------------------------------------------- precompute
+    @torch.no_grad()
+    def precompute(self):
+        """use_pp pass (reference train.py:170-211): one FULL (unsampled)
+        boundary feature exchange + first-layer aggregation, after which
+        training layer 0 degenerates to a GEMM."""
+        args, ctx = self.args, self.ctx
+        st = ctx.full_state()
+        send = pack_rows_raw(self.feat, st.pack_idx, None)
+        recv = torch.empty(sum(st.recv_counts), self.feat.shape[1],
+                           dtype=self.feat.dtype, device=self.feat.device)
+        all_to_all_rows(recv, send, st.recv_counts, st.send_counts)
+        if args.model == "gcn":
+            out = spmm_sum_raw(ctx.indptr, ctx.indices, self.feat,
+                               ctx.out_norm_inv, ctx.in_norm_inv)
+            spmm_sum_raw(st.halo_fwd_indptr, st.halo_fwd_indices, recv,
+                         src_scale=st.halo_out_norm_inv,
+                         dst_scale=ctx.in_norm_inv, out=out)
+            self.feat = out
+        elif args.model == "graphsage":
+            mean = spmm_sum_raw(ctx.indptr, ctx.indices, self.feat,
+                                None, ctx.in_deg_inv)
+            spmm_sum_raw(st.halo_fwd_indptr, st.halo_fwd_indices, recv,
+                         src_scale=None, dst_scale=ctx.in_deg_inv, out=mean)
+            self.feat = torch.cat((self.feat, mean), dim=1)
+        elif args.model == "gat":
+            self.halo_feat0 = recv   # keep raw full-halo features
+        else:
+            raise ValueError(args.model)
+
+
+def _forward(model, state: RankState, feat):
+    if state.args.model == "gat":
+        return model(state.ctx, feat, halo_feat0=state.halo_feat0)
+    return model(state.ctx, feat)
+
+
+# -------------------------------------------------------------- evaluation
+
+def _accuracy(logits: torch.Tensor, labels: torch.Tensor, multilabel: bool) -> float:
+    if multilabel:
+        pred = (logits > 0).float()
+        tp = (pred * labels).sum()
+        fp = (pred * (1 - labels)).sum()
+        fn = ((1 - pred) * labels).sum()
+        return float(2 * tp / (2 * tp + fp + fn + 1e-12))
+    return float((logits.argmax(1) == labels).float().mean())
+
+
+class Evaluator:
+    """Full-graph inference on CPU (rank 0), reference train.py:13-61.
+
+    Regenerates the deterministic synthetic dataset instead of loading it
+    from disk; transductive: one pass over the full graph scoring val+test;
+    inductive: val on the train∪val subgraph, test on the full graph."""
+
+    def __init__(self, args):
+        self.args = args
+        self.multilabel = None
+        g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
+        self.multilabel = g.multilabel
+        self._graphs = {}
+        if args.inductive:
+            self._graphs["val"] = g.subgraph(g.train_mask | g.val_mask)
+            self._graphs["test"] = g
+        else:
+            self._graphs["full"] = g
+
+    def _ctx_and_tensors(self, g):
+        ctx = GraphContext.for_full_graph(
+            torch.from_numpy(g.adj_in.indptr), torch.from_numpy(g.adj_in.indices),
+            torch.from_numpy(g.in_deg), torch.from_numpy(g.out_deg), "cpu")
+        feat = torch.from_numpy(g.feat)
+        label = torch.from_numpy(g.label)
+        return ctx, feat, label, g
+
+    @torch.no_grad()
+    def evaluate(self, model_cpu) -> dict:
+        model_cpu.eval()
+        out = {}
+        if self.args.inductive:
+            ctx, feat, label, g = self._ctx_and_tensors(self._graphs["val"])
+            logits = model_cpu(ctx, feat)
+            vm = torch.from_numpy(g.val_mask)
+            out["val"] = _accuracy(logits[vm], label[vm], self.multilabel)
+        else:
+            ctx, feat, label, g = self._ctx_and_tensors(self._graphs["full"])
+            logits = model_cpu(ctx, feat)
+            vm = torch.from_numpy(g.val_mask)
+            tm = torch.from_numpy(g.test_mask)
+            out["val"] = _accuracy(logits[vm], label[vm], self.multilabel)
+            out["test"] = _accuracy(logits[tm], label[tm], self.multilabel)
+        return out
+
+    @torch.no_grad()
+    def evaluate_test(self, model_cpu) -> float:
+        model_cpu.eval()
+        key = "test" if self.args.inductive else "full"
+        ctx, feat, label, g = self._ctx_and_tensors(self._graphs[key])
+        logits = model_cpu(ctx, feat)
+        tm = torch.from_numpy(g.test_mask)
+        return _accuracy(logits[tm], label[tm], self.multilabel)
+
+
+# --------------------------------------------------------------- the loop
+
+def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
+    """One training process (= one partition = one GPU). Returns summary
+    stats (epoch-time mean, loss, accuracies) for tests/bench."""
+    rank, world = init_distributed(args.backend, rank, world_size,
+                                   args.master_addr, args.port)
+    if args.device == "auto":
+        device = f"cuda:{rank % torch.cuda.device_count()}" \
+            if torch.cuda.is_available() else "cpu"
+    else:
+        device = args.device
+    if str(device).startswith("cuda"):
+        torch.cuda.set_device(torch.device(device))
+
+    name = graph_name_of(args)
+    part = load_partition(args.partition_dir, name, rank)
+    assert part.n_parts == world, (part.n_parts, world)
+    if args.fix_seed:
+        torch.manual_seed(args.seed)
+
+    state = RankState(part, args, device)
+    state.plan.set_epoch(0)
+    model = create_model(args, n_feat=state.feat.shape[1],
+                         n_class=int(part.meta["n_class"]),
+                         train_size=state.n_train_global).to(device)
+    if args.use_pp or args.model == "gat":
+        state.precompute()
+
+    # broadcast initial weights so all ranks start identical (the reference
+    # relies on --fix-seed for this, main.py:13-16; we make it robust)
+    if world > 1:
+        for p in model.parameters():
+            dist.broadcast(p.data, src=0)
+
+    multilabel = bool(part.meta.get("multilabel", False))
+    if multilabel:
+        loss_fcn = torch.nn.BCEWithLogitsLoss(reduction="sum")
+        labels_train = state.label[state.train_mask].float()
+    else:
+        loss_fcn = torch.nn.CrossEntropyLoss(reduction="sum")
+        labels_train = state.label[state.train_mask].long()
+
+    reducer = GradReducer(model, state.n_train_global,
+                          bucket_bytes=args.bucket_mb << 20)
+    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr,
+                                 weight_decay=args.weight_decay)
+
+    evaluator = None
+    pool = None
+    pending = None
+    best_val, best_state = -1.0, None
+    if args.eval and rank == 0:
+        evaluator = Evaluator(args)
+        pool = ThreadPoolExecutor(max_workers=1)
+        os.makedirs("checkpoint", exist_ok=True)
+        os.makedirs("results", exist_ok=True)
+
+    train_dur, comm_dur, reduce_dur = [], [], []
+    loss_history: list[float] = []
+    loss_val = float("nan")
+    cuda = str(device).startswith("cuda")
+
+    for epoch in range(args.n_epochs):
+        t0 = time.perf_counter()
+        state.plan.set_epoch(epoch)
+        model.train()
+        logits = _forward(model, state, state.feat)
+        loss = loss_fcn(logits[state.train_mask], labels_train)
+        reducer.zero_grad()
+        loss.backward()
+        t_red = time.perf_counter()
+        reducer.reduce()
+        reducer.synchronize()
+        reduce_dur.append(time.perf_counter() - t_red)
+        optimizer.step()
+        if cuda:
+            torch.cuda.synchronize()
+        if epoch >= 5:
+            train_dur.append(time.perf_counter() - t0)
+        comm_dur.append(comm_timer.tot_time())
+        comm_timer.clear()
+        loss_val = loss.item()
+        loss_history.append(loss_val)
+
+        if (epoch + 1) % args.log_every == 0:
+            print(f"Process {rank:03d} | Epoch {epoch:05d} | "
+                  f"Time(s) {np.mean(train_dur) if train_dur else 0:.4f} | "
+                  f"Comm(s) {np.mean(comm_dur):.4f} | "
+                  f"Reduce(s) {np.mean(reduce_dur):.4f} | "
+                  f"Loss {loss_val / max(state.part_train, 1):.4f}", flush=True)
+
+        if evaluator is not None and (epoch + 1) % args.log_every == 0:
+            torch.save(model.state_dict(),
+                       f"checkpoint/{name}_p{args.sampling_rate:.2f}_{epoch}.pth.tar")
+            if pending is not None:
+                res, snap = pending.result()
+                if res["val"] > best_val:
+                    best_val, best_state = res["val"], snap
+            model_cpu = copy.deepcopy(model).cpu()
+            snap = copy.deepcopy(model_cpu.state_dict())
+            pending = pool.submit(lambda m=model_cpu, s=snap:
+                                  (evaluator.evaluate(m), s))
+
+    summary = {"rank": rank, "epoch_time": float(np.mean(train_dur)) if train_dur
+               else float("nan"),
+               "comm_time": float(np.mean(comm_dur)) if comm_dur else 0.0,
+               "loss": loss_val / max(state.part_train, 1),
+               "loss_history": loss_history}
+    if evaluator is not None:
+        if pending is not None:
+            res, snap = pending.result()
+            if res["val"] > best_val:
+                best_val, best_state = res["val"], snap
+        if best_state is not None:
+            torch.save(best_state, f"checkpoint/{name}_final.pth.tar")
+            # rebuild with the ORIGINAL feature width (precompute may have
+            # widened state.feat; eval consumes raw features)
+            final = create_model(args, n_feat=int(part.meta["n_feat"]),
+                                 n_class=int(part.meta["n_class"]),
+                                 train_size=state.n_train_global)
+            final.load_state_dict(best_state)
+            test_acc = evaluator.evaluate_test(final)
+            summary["val_acc"] = best_val
+            summary["test_acc"] = test_acc
+            with open(f"results/{args.dataset}_n{args.n_partitions}"
+                      f"_p{args.sampling_rate:.2f}.txt", "a") as f:
+                f.write(f"val={best_val:.4f} test={test_acc:.4f}\n")
+            print(f"Max Validation Accuracy {best_val:.4f} | "
+                  f"Test Accuracy {test_acc:.4f}", flush=True)
+        if pool is not None:
+            pool.shutdown()
+    return summary

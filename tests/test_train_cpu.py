@@ -1,0 +1,149 @@
+"""End-to-end CPU (gloo) training tests — the plumbing oracle.
+
+The strongest check: with sampling-rate 1.0 and dropout 0, distributed
+partition-parallel training must match single-process full-graph training
+EXACTLY (same loss trajectory up to fp accumulation order) — this
+validates boundary discovery, the halo exchange, aggregation scales,
+gradient reduction and the optimizer wiring all at once. This is
+BASELINE.json config 1 (CPU/gloo plumbing, no GPU).
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from bnsgcn_amd.runtime.config import create_parser, graph_name_of
+from bnsgcn_amd.runtime.trainer import prepare_partitions, run
+
+from util_dist import run_dist
+
+
+def make_args(tmp_path, **kw):
+    args = create_parser().parse_args([])
+    args.dataset = "tiny"
+    args.n_hidden = 16
+    args.n_layers = 2
+    args.n_epochs = 12
+    args.dropout = 0.0
+    args.fix_seed = True
+    args.seed = 7
+    args.eval = False
+    args.backend = "gloo"
+    args.device = "cpu"
+    args.log_every = 100
+    args.partition_dir = str(tmp_path / "partition")
+    for k, v in kw.items():
+        setattr(args, k, v)
+    args.graph_name = graph_name_of(args)
+    return args
+
+
+def _train(rank, world, args):
+    return run(args, rank=rank, world_size=world)
+
+
+def _run_config(tmp_path, world, **kw):
+    args = make_args(tmp_path, n_partitions=world, **kw)
+    prepare_partitions(args)
+    # always fork (even world=1): initializing gloo in the pytest parent
+    # would poison later forked children with copied gloo threads
+    return run_dist(world, _train, (args,))
+
+
+@pytest.mark.parametrize("model,use_pp", [("graphsage", False),
+                                          ("graphsage", True),
+                                          ("gcn", False),
+                                          ("gcn", True)])
+def test_dist_matches_single_at_full_rate(tmp_path, model, use_pp):
+    single = _run_config(tmp_path / "s", 1, model=model, use_pp=use_pp,
+                         sampling_rate=1.0)
+    multi = _run_config(tmp_path / "m", 2, model=model, use_pp=use_pp,
+                        sampling_rate=1.0)
+    lh_single = np.array(single[0]["loss_history"])
+    lh_multi = np.array(multi[0]["loss_history"]) + np.array(multi[1]["loss_history"])
+    np.testing.assert_allclose(lh_multi, lh_single, rtol=2e-3, atol=1e-3)
+    # training actually learns
+    assert lh_single[-1] < lh_single[0]
+
+
+def test_dist_matches_single_p4(tmp_path):
+    single = _run_config(tmp_path / "s", 1, model="graphsage", sampling_rate=1.0)
+    multi = _run_config(tmp_path / "m", 4, model="graphsage", sampling_rate=1.0,
+                        partition_method="random")
+    lh_single = np.array(single[0]["loss_history"])
+    lh_multi = sum(np.array(m["loss_history"]) for m in multi)
+    np.testing.assert_allclose(lh_multi, lh_single, rtol=5e-3, atol=1e-3)
+
+
+@pytest.mark.parametrize("model", ["graphsage", "gcn"])
+def test_sampled_training_learns(tmp_path, model):
+    multi = _run_config(tmp_path, 2, model=model, sampling_rate=0.3,
+                        use_pp=True, n_epochs=50, dropout=0.1)
+    for m in multi:
+        lh = np.array(m["loss_history"])
+        assert lh[-1] < lh[0] * 0.9
+
+
+def test_gat_trains(tmp_path):
+    multi = _run_config(tmp_path, 2, model="gat", heads=2, n_hidden=8,
+                        sampling_rate=0.5, n_epochs=20)
+    for m in multi:
+        lh = np.array(m["loss_history"])
+        assert np.isfinite(lh).all()
+        assert lh[-1] < lh[0]
+
+
+def test_gat_dist_matches_single_at_full_rate(tmp_path):
+    single = _run_config(tmp_path / "s", 1, model="gat", heads=2, n_hidden=8,
+                         sampling_rate=1.0, n_epochs=10)
+    multi = _run_config(tmp_path / "m", 2, model="gat", heads=2, n_hidden=8,
+                        sampling_rate=1.0, n_epochs=10)
+    lh_single = np.array(single[0]["loss_history"])
+    lh_multi = np.array(multi[0]["loss_history"]) + np.array(multi[1]["loss_history"])
+    np.testing.assert_allclose(lh_multi, lh_single, rtol=5e-3, atol=1e-3)
+
+
+def test_multilabel_bce(tmp_path):
+    multi = _run_config(tmp_path, 2, dataset="tiny-ml", model="graphsage",
+                        sampling_rate=0.5, use_pp=True, n_epochs=15)
+    for m in multi:
+        lh = np.array(m["loss_history"])
+        assert lh[-1] < lh[0]
+
+
+def test_sampling_rate_zero(tmp_path):
+    multi = _run_config(tmp_path, 2, model="graphsage", sampling_rate=0.0,
+                        use_pp=True, n_epochs=8)
+    for m in multi:
+        assert np.isfinite(m["loss_history"]).all()
+
+
+def test_inductive_mode(tmp_path):
+    multi = _run_config(tmp_path, 2, model="graphsage", sampling_rate=0.5,
+                        use_pp=True, inductive=True, n_epochs=8)
+    for m in multi:
+        assert np.isfinite(m["loss_history"]).all()
+
+
+def test_eval_and_checkpoint(tmp_path):
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        multi = _run_config(tmp_path, 2, model="graphsage", sampling_rate=0.5,
+                            use_pp=True, n_epochs=10, log_every=5, eval=True)
+        assert "test_acc" in multi[0]
+        assert 0.0 <= multi[0]["test_acc"] <= 1.0
+        name = "tiny-2-metis-vol-trans"
+        assert os.path.exists(f"checkpoint/{name}_p0.50_4.pth.tar")
+        assert os.path.exists(f"checkpoint/{name}_final.pth.tar")
+        assert os.path.exists("results/tiny_n2_p0.50.txt")
+    finally:
+        os.chdir(cwd)
+
+
+def test_syncbn_norm(tmp_path):
+    multi = _run_config(tmp_path, 2, model="graphsage", sampling_rate=1.0,
+                        norm="batch", inductive=True, use_pp=True, n_epochs=8)
+    for m in multi:
+        assert np.isfinite(m["loss_history"]).all()
