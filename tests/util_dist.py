@@ -30,9 +30,13 @@ def _entry(rank, world, port, fn, fargs, q):
             dist.destroy_process_group()
 
 
-def run_dist(world: int, fn, fargs=(), timeout: float = 180.0) -> list:
-    """Returns [result_rank0, result_rank1, ...]; raises on any failure."""
-    ctx = pymp.get_context("fork")
+def run_dist(world: int, fn, fargs=(), timeout: float = 300.0) -> list:
+    """Returns [result_rank0, result_rank1, ...]; raises on any failure.
+
+    Uses the spawn start method: fork is unsafe once the pytest parent has
+    run torch CPU ops (OpenMP pools) or initialized gloo — forked children
+    inherit locked thread state and deadlock."""
+    ctx = pymp.get_context("spawn")
     q = ctx.Queue()
     port = free_port()
     ps = [ctx.Process(target=_entry, args=(r, world, port, fn, fargs, q))
