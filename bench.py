@@ -1,0 +1,194 @@
+"""Flagship benchmark — the driver contract.
+
+Measures the BASELINE.json headline metric: epoch time + throughput
+(edges/s) for Reddit-shaped GraphSAGE 3-layer h=256, sampling-rate=0.1,
+use_pp, on N partitions = N GPUs (one process per GPU over RCCL/xGMI).
+
+  python bench.py --gpus N --steps K --warmup W
+  (N>1 is launched by the driver via torch.distributed.run, one rank/GPU)
+
+A "step" is one full training epoch over the partitioned graph (forward,
+loss, backward, gradient all-reduce, Adam step) — identical work to the
+reference's epoch loop (reference train.py:385-425). Data is synthetic of
+the named shape (232,965 nodes / ~114.8M directed edges incl self-loops,
+602 features, 41 classes — graph/synthetic.py) with random-init weights;
+compute dtype fp32 = the reference's dtype.
+
+Scaling is STRONG: the same fixed graph is partitioned across N GPUs.
+
+value       = full-graph edges / epoch-time  (whole-job aggregate)
+vs_baseline = value / (E_ref / 0.3578s), the reference README's measured
+              epoch time (note: the README quotes a 4-layer,
+              2-GPU, NVIDIA run of the same dataset+rate — the closest
+              published number; BASELINE.json names the 3-layer/8-part
+              config measured here).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--dataset", type=str, default="reddit")
+    p.add_argument("--model", type=str, default="graphsage")
+    p.add_argument("--n-layers", type=int, default=3)
+    p.add_argument("--n-hidden", type=int, default=256)
+    p.add_argument("--heads", type=int, default=4)
+    p.add_argument("--sampling-rate", type=float, default=0.1)
+    p.add_argument("--partition-method", type=str, default="metis")
+    p.add_argument("--data-scale", type=float, default=1.0)
+    p.add_argument("--partition-dir", type=str, default="bench_partition")
+    p.add_argument("--device", type=str, default="auto")
+    return p.parse_args()
+
+
+def main():
+    a = parse()
+    from bnsgcn_amd.runtime.config import create_parser, graph_name_of
+    from bnsgcn_amd.runtime.trainer import prepare_partitions, RankState, _forward
+    from bnsgcn_amd.graph import load_partition, load_meta
+    from bnsgcn_amd.models.models import create_model
+    from bnsgcn_amd.parallel import GradReducer, init_distributed
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", a.gpus))
+    assert world == a.gpus, f"WORLD_SIZE {world} != --gpus {a.gpus}"
+
+    args = create_parser().parse_args([])
+    args.dataset = a.dataset
+    args.model = a.model
+    args.n_layers = a.n_layers
+    args.n_hidden = a.n_hidden
+    args.heads = a.heads
+    args.sampling_rate = a.sampling_rate
+    args.n_partitions = world
+    args.partition_method = a.partition_method
+    args.data_scale = a.data_scale
+    args.partition_dir = a.partition_dir
+    args.use_pp = True
+    args.eval = False
+    args.fix_seed = True
+    args.seed = 0
+    args.device = a.device
+    args.graph_name = graph_name_of(args)
+
+    cuda = torch.cuda.is_available() and a.device != "cpu"
+    if world > 1:
+        rank, world = init_distributed("nccl" if cuda else "gloo")
+    if rank == 0:
+        args.skip_partition = True   # reuse an existing store for this config
+        prepare_partitions(args)
+    if world > 1:
+        dist.barrier()
+
+    if a.device == "auto":
+        device = f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}" if cuda else "cpu"
+    else:
+        device = a.device
+    if cuda:
+        torch.cuda.set_device(torch.device(device))
+
+    part = load_partition(args.partition_dir, args.graph_name, rank)
+    meta = part.meta
+    torch.manual_seed(args.seed)
+    state = RankState(part, args, device)
+    state.plan.set_epoch(0)
+    model = create_model(args, n_feat=int(meta["n_feat"]),
+                         n_class=int(meta["n_class"]),
+                         train_size=int(meta["n_train"])).to(device)
+    state.precompute()
+    if world > 1:
+        for prm in model.parameters():
+            dist.broadcast(prm.data, src=0)
+
+    multilabel = bool(meta.get("multilabel", False))
+    if multilabel:
+        loss_fcn = torch.nn.BCEWithLogitsLoss(reduction="sum")
+        labels_train = state.label[state.train_mask].float()
+    else:
+        loss_fcn = torch.nn.CrossEntropyLoss(reduction="sum")
+        labels_train = state.label[state.train_mask].long()
+    reducer = GradReducer(model, int(meta["n_train"]))
+    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr)
+
+    def step(epoch: int):
+        state.plan.set_epoch(epoch)
+        model.train()
+        logits = _forward(model, state, state.feat)
+        loss = loss_fcn(logits[state.train_mask], labels_train)
+        reducer.zero_grad()
+        loss.backward()
+        reducer.reduce()
+        reducer.synchronize()
+        optimizer.step()
+        return loss
+
+    for e in range(a.warmup):
+        step(e)
+
+    if cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for e in range(a.steps):
+        step(a.warmup + e)
+    if cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed])
+    if world > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t[0])
+
+    if rank == 0:
+        n_edges = int(meta.get("full_n_edges", meta["n_edges"]))
+        epoch_s = elapsed / a.steps
+        value = n_edges / epoch_s
+        ref_epoch_s = 0.3578          # BASELINE.md (README.md:94-95)
+        baseline = n_edges / ref_epoch_s if a.dataset == "reddit" and \
+            a.data_scale == 1.0 else None
+        out = {
+            "metric": "training throughput (full-graph edges/s), Reddit-shaped "
+                      "GraphSAGE 3-layer h=256, sampling-rate=0.1",
+            "value": value,
+            "unit": "edges/s",
+            "n_gpus": world,
+            "steps": a.steps,
+            "warmup": a.warmup,
+            "ms_per_step": epoch_s * 1e3,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": (value / baseline) if baseline else None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {"model": a.model, "dataset": a.dataset,
+                       "n_layers": a.n_layers, "n_hidden": a.n_hidden,
+                       "sampling_rate": a.sampling_rate,
+                       "n_nodes": int(meta.get("full_n_nodes", meta["n_nodes"])),
+                       "n_edges": n_edges, "use_pp": True,
+                       "partition": a.partition_method,
+                       "parallelism": f"partition-parallel p{world}",
+                       "epoch_time_s": epoch_s},
+        }
+        print(json.dumps(out), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -45,7 +45,13 @@ class CSR:
         counts = np.bincount(dst, minlength=n_rows).astype(np.int64)
         indptr = np.zeros(n_rows + 1, dtype=np.int64)
         np.cumsum(counts, out=indptr[1:])
-        order = np.argsort(dst, kind="stable")
+        if len(dst) > 4_000_000:
+            # torch's parallel stable sort is ~2x numpy on big arrays
+            import torch
+            order = torch.argsort(torch.from_numpy(np.ascontiguousarray(dst)),
+                                  stable=True).numpy()
+        else:
+            order = np.argsort(dst, kind="stable")
         indices = src[order].astype(np.int32)
         csr = CSR(indptr, indices, n_cols)
         if sort_cols:

@@ -51,39 +51,44 @@ def partition_graph(g: Graph, n_parts: int, method: str = "metis", seed: int = 0
         inner_local[nodes] = np.arange(len(nodes))
         inner_lists.append(nodes)
 
-    src, dst = g.adj_in.to_edges()                  # int32 arrays
+    src, dst = g.adj_in.to_edges()                  # dst is NONDECREASING (CSR order)
     spart = part[src]
     dpart = part[dst]
 
     parts: list[Partition] = []
     boundary_all: list[list[np.ndarray]] = [[None] * n_parts for _ in range(n_parts)]
 
+    from .csr import CSR
     for p in range(n_parts):
         nodes = inner_lists[p]
         n_inner = len(nodes)
-        em = dpart == p
-        e_src, e_dst = src[em], dst[em]
-        e_spart = spart[em]
+        # flatnonzero keeps ascending edge order, so dst stays nondecreasing
+        # inside the bucket — per-partition CSRs need NO re-sort.
+        eidx = np.flatnonzero(dpart == p)
+        e_src, e_dst = src[eidx], dst[eidx]
+        e_spart = spart[eidx]
         dst_local = inner_local[e_dst].astype(np.int32)
 
-        # inner->inner edges
+        # inner->inner edges: rows (dst_local) already nondecreasing
         im = e_spart == p
-        from .csr import CSR
-        inner_csr = CSR.from_edges(inner_local[e_src[im]].astype(np.int32),
-                                   dst_local[im], n_inner, n_inner, sort_cols=True)
+        i_dst = dst_local[im]
+        counts = np.bincount(i_dst, minlength=n_inner).astype(np.int64)
+        inner_indptr = np.zeros(n_inner + 1, dtype=np.int64)
+        np.cumsum(counts, out=inner_indptr[1:])
+        inner_csr = CSR(inner_indptr, inner_local[e_src[im]].astype(np.int32),
+                        n_inner)
 
         # halo edges grouped by (owner, owner-local id)
         hm = ~im
         h_owner = e_spart[hm]
-        h_ol = inner_local[e_src[hm]].astype(np.int32)   # owner-local src id
+        h_ol = inner_local[e_src[hm]].astype(np.int64)   # owner-local src id
         h_dst = dst_local[hm]
         # unique halo rows in (owner, owner_local) order == peer-major sorted
         key = h_owner.astype(np.int64) * n + h_ol
         uniq, inv = np.unique(key, return_inverse=True)
         halo_part = (uniq // n).astype(np.int32)
         halo_ol = (uniq % n).astype(np.int32)
-        halo_csr = CSR.from_edges(h_dst, inv.astype(np.int64), len(uniq), n_inner,
-                                  sort_cols=True)
+        halo_csr = CSR.from_edges(h_dst, inv.astype(np.int64), len(uniq), n_inner)
 
         # halo degrees come straight from the full graph (global view)
         halo_global = np.empty(len(uniq), dtype=np.int64)

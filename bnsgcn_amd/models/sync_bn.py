@@ -27,7 +27,11 @@ class _SyncBNFunc(Function):
         if not training:
             mean, var = running_mean, running_var
         else:
-            stats = torch.stack((x.sum(0), (x * x).sum(0)))
+            from ..ops._ext import use_hip, get_ext
+            if use_hip(x):
+                stats = get_ext().syncbn_stats(x.contiguous())
+            else:
+                stats = torch.stack((x.sum(0), (x * x).sum(0)))
             _maybe_all_reduce(stats)
             mean = stats[0] / whole_size
             var = (stats[1] - mean * stats[0]) / whole_size

@@ -1,0 +1,623 @@
+// bnsgcn_amd gfx950 kernel library — batch 1 (the GCN/SAGE hot path).
+//
+// Hand-written HIP for CDNA4 (MI355X): wave64, float4-vectorized HBM
+// access, MFMA (v_mfma_f32_16x16x4_f32) for the dense GEMMs. No CUDA
+// compatibility paths, no Triton. Reference-kernel parity map in
+// spmm_sum=K1/K2/K3+K18, pack_rows=K13,
+// scatter_add_rows=K14, philox_keys=K16, gemm_*=K6, syncbn_stats=K12.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/util/Optional.h>
+#include <algorithm>
+#include "common.h"
+
+namespace {
+
+// ============================== SpMM ===================================
+// out[r, :] (+)= dst_scale[r] * sum_{e in row r} src_scale[col_e] * x[col_e, :]
+// One wave per output row (4 rows / 256-thread block), grid-stride over
+// rows. Feature dim covered by the 64 lanes: float4 path when F % 4 == 0
+// (64 lanes x 16 B = 1 KiB per neighbor row read — full coalescing),
+// scalar path otherwise. Rows are independent -> no atomics; the backward
+// pass reuses this kernel on the precomputed transposed CSR.
+
+template <bool ACC>
+__global__ void spmm_sum_vec4_kernel(
+    const int64_t* __restrict__ indptr, const int32_t* __restrict__ indices,
+    const float* __restrict__ x, const float* __restrict__ src_scale,
+    const float* __restrict__ dst_scale, float* __restrict__ out,
+    int n_rows, int f4 /* F/4 */) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const float4* __restrict__ x4 = reinterpret_cast<const float4*>(x);
+  float4* __restrict__ out4 = reinterpret_cast<float4*>(out);
+
+  for (int r = wave; r < n_rows; r += n_waves) {
+    const int64_t beg = indptr[r], end = indptr[r + 1];
+    const float ds = dst_scale ? dst_scale[r] : 1.0f;
+    for (int f = lane; f < f4; f += WAVE) {
+      float4 acc = {0.f, 0.f, 0.f, 0.f};
+      for (int64_t e = beg; e < end; ++e) {
+        const int c = indices[e];
+        const float s = src_scale ? src_scale[c] : 1.0f;
+        const float4 v = x4[(int64_t)c * f4 + f];
+        acc.x += s * v.x; acc.y += s * v.y;
+        acc.z += s * v.z; acc.w += s * v.w;
+      }
+      const int64_t o = (int64_t)r * f4 + f;
+      if (ACC) {
+        float4 prev = out4[o];
+        prev.x += ds * acc.x; prev.y += ds * acc.y;
+        prev.z += ds * acc.z; prev.w += ds * acc.w;
+        out4[o] = prev;
+      } else {
+        out4[o] = make_float4(ds * acc.x, ds * acc.y, ds * acc.z, ds * acc.w);
+      }
+    }
+  }
+}
+
+template <bool ACC>
+__global__ void spmm_sum_scalar_kernel(
+    const int64_t* __restrict__ indptr, const int32_t* __restrict__ indices,
+    const float* __restrict__ x, const float* __restrict__ src_scale,
+    const float* __restrict__ dst_scale, float* __restrict__ out,
+    int n_rows, int F) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int r = wave; r < n_rows; r += n_waves) {
+    const int64_t beg = indptr[r], end = indptr[r + 1];
+    const float ds = dst_scale ? dst_scale[r] : 1.0f;
+    for (int f = lane; f < F; f += WAVE) {
+      float acc = 0.f;
+      for (int64_t e = beg; e < end; ++e) {
+        const int c = indices[e];
+        const float s = src_scale ? src_scale[c] : 1.0f;
+        acc += s * x[(int64_t)c * F + f];
+      }
+      const int64_t o = (int64_t)r * F + f;
+      if (ACC) out[o] += ds * acc; else out[o] = ds * acc;
+    }
+  }
+}
+
+// ========================= pack / scatter ==============================
+// pack: out[r, :] = scale[r] * x[idx[r], :]        (send-side gather, K13)
+// scatter: out[idx[r], :] += scale[r] * src[r, :]  (grad unpack, K14).
+// Scatter destinations can repeat ACROSS peers (a node bordering several
+// partitions) -> atomicAdd (fp32, device scope).
+
+__global__ void pack_rows_kernel(const float* __restrict__ x,
+                                 const int64_t* __restrict__ idx,
+                                 const float* __restrict__ scale,
+                                 float* __restrict__ out, int n, int F) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (int64_t)n * F;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = t; i < total; i += stride) {
+    const int r = i / F, f = i - (int64_t)r * F;
+    const float s = scale ? scale[r] : 1.0f;
+    out[i] = s * x[idx[r] * F + f];
+  }
+}
+
+__global__ void scatter_add_rows_kernel(float* __restrict__ out,
+                                        const int64_t* __restrict__ idx,
+                                        const float* __restrict__ src,
+                                        const float* __restrict__ scale,
+                                        int n, int F) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (int64_t)n * F;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = t; i < total; i += stride) {
+    const int r = i / F, f = i - (int64_t)r * F;
+    const float s = scale ? scale[r] : 1.0f;
+    atomicAdd(&out[idx[r] * F + f], s * src[i]);
+  }
+}
+
+// ============================ Philox keys ==============================
+// Device-side BNS sampling keys — bitwise-identical to ops/philox.py
+// (numpy); the sorted-top-s selection runs as torch.sort on device.
+
+__global__ void philox_keys_kernel(int64_t* __restrict__ out, int n,
+                                   uint32_t seed_lo, uint32_t seed_hi,
+                                   uint32_t epoch, uint32_t sd) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  P4 r = philox4x32((uint32_t)i, epoch, sd, 0x424E5347u, seed_lo, seed_hi);
+  uint64_t key = ((uint64_t)r.c0 << 32) | r.c1;
+  out[i] = (int64_t)(key & 0x7FFFFFFFFFFFFFFFull);
+}
+
+// =========================== SyncBN stats ==============================
+// Column sums: out[0, f] = sum_r x[r, f]; out[1, f] = sum_r x[r, f]^2.
+// Lane-per-column (coalesced row-major reads), one block per (col-chunk,
+// row-chunk), block-partial then atomicAdd — few atomics per column.
+
+__global__ void syncbn_stats_kernel(const float* __restrict__ x,
+                                    float* __restrict__ out,
+                                    int64_t n, int F, int row_chunks) {
+  const int f = blockIdx.x * blockDim.x + threadIdx.x;
+  if (f >= F) return;
+  const int64_t rows_per = (n + row_chunks - 1) / row_chunks;
+  const int64_t r0 = blockIdx.y * rows_per;
+  const int64_t r1 = (r0 + rows_per < n) ? r0 + rows_per : n;
+  float s = 0.f, s2 = 0.f;
+  for (int64_t r = r0; r < r1; ++r) {
+    const float v = x[r * F + f];
+    s += v; s2 += v * v;
+  }
+  if (row_chunks == 1) { out[f] = s; out[F + f] = s2; }
+  else { atomicAdd(&out[f], s); atomicAdd(&out[F + f], s2); }
+}
+
+// ============================ fp32 MFMA GEMM ===========================
+// C[M,N] = A' @ B' (+bias), exact fp32 via v_mfma_f32_16x16x4_f32 (the
+// CDNA4 f32-input MFMA: A lane map A[i=l&15][k=l>>4], B[k=l>>4][j=l&15],
+// C/D col=l&15, row=(l>>4)*4+reg — cdna_hip_programming.md §3).
+// Layout-generic via element strides (handles NT/NN/TN from one kernel);
+// 64x64 block tile, 4 waves each owning a 32x32 quadrant (2x2 fragments),
+// BK=16 LDS-staged with +1-float row pad against bank conflicts.
+// These GEMMs are tall-skinny (M up to ~500k, N,K in 41..1204) and
+// memory-bound; correctness + coalesced staging matter, peak MFMA does not.
+
+#define BM 64
+#define BN 64
+#define BK 16
+
+__global__ __launch_bounds__(256)
+void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
+                     const float* __restrict__ B, int64_t sBk, int64_t sBn,
+                     const float* __restrict__ bias, float* __restrict__ C,
+                     int M, int N, int K) {
+  __shared__ float As[BM][BK + 1];
+  __shared__ float Bs[BK][BN + 1];
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;          // 4 waves: quadrant (wr, wc)
+  const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
+
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  f32x4 acc[2][2] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // cooperative staging: 256 threads, A: 64x16 (4 elems/thread),
+    // B: 16x64 (4 elems/thread); bounds-checked (zero-fill).
+    for (int t = tid; t < BM * BK; t += 256) {
+      const int i = t / BK, k = t % BK;
+      const int gm = m0 + i, gk = k0 + k;
+      As[i][k] = (gm < M && gk < K) ? A[gm * sAm + gk * sAk] : 0.f;
+    }
+    for (int t = tid; t < BK * BN; t += 256) {
+      const int k = t / BN, j = t % BN;
+      const int gk = k0 + k, gn = n0 + j;
+      Bs[k][j] = (gk < K && gn < N) ? B[gk * sBk + gn * sBn] : 0.f;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int i_l = lane & 15, k_l = lane >> 4;   // fragment lane map
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi) {
+        const float a = As[wr + fi * 16 + i_l][kk + k_l];
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj) {
+          const float b = Bs[kk + k_l][wc + fj * 16 + i_l];
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b,
+                                                             acc[fi][fj], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C/D map col=l&15, row=(l>>4)*4+reg
+  const int j_l = lane & 15, r_l = lane >> 4;
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gm = m0 + wr + fi * 16 + r_l * 4 + r;
+        const int gn = n0 + wc + fj * 16 + j_l;
+        if (gm < M && gn < N) {
+          float v = acc[fi][fj][r];
+          if (bias) v += bias[gn];
+          C[(int64_t)gm * N + gn] = v;
+        }
+      }
+}
+
+// ============================ GAT kernel set ===========================
+// SDDMM u_add_v, segmented edge-softmax fwd/bwd, edge-weighted SpMM and
+// per-edge dot (reference K4/K5 — DGL GATConv internals). Edge tensors
+// are CSR-ordered (grouped by destination row), so softmax segments are
+// contiguous: one wave per (row, head) with shfl-based reductions.
+
+DEV_INLINE float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+DEV_INLINE float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
+  return v;
+}
+
+// logits[e,h] = el[col_e,h] + er[row_e,h] — edge-parallel, row via the
+// contiguous CSR segment (wave per row, lanes stride edges*heads).
+__global__ void sddmm_add_kernel(const int64_t* __restrict__ indptr,
+                                 const int32_t* __restrict__ indices,
+                                 const float* __restrict__ el,
+                                 const float* __restrict__ er,
+                                 float* __restrict__ out, int n_rows, int H) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int r = wave; r < n_rows; r += n_waves) {
+    const int64_t beg = indptr[r], end = indptr[r + 1];
+    const int64_t cnt = (end - beg) * H;
+    for (int64_t t = lane; t < cnt; t += WAVE) {
+      const int64_t e = beg + t / H;
+      const int h = t % H;
+      out[e * H + h] = el[(int64_t)indices[e] * H + h] + er[(int64_t)r * H + h];
+    }
+  }
+}
+
+// alpha[e,h] = softmax over each row's edge segment (numerically stable).
+__global__ void segment_softmax_kernel(const int64_t* __restrict__ indptr,
+                                       const float* __restrict__ logits,
+                                       float* __restrict__ alpha,
+                                       int n_rows, int H) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int rh = wave; rh < n_rows * H; rh += n_waves) {
+    const int r = rh / H, h = rh % H;
+    const int64_t beg = indptr[r], end = indptr[r + 1];
+    if (beg == end) continue;
+    float m = -INFINITY;
+    for (int64_t e = beg + lane; e < end; e += WAVE)
+      m = fmaxf(m, logits[e * H + h]);
+    m = wave_reduce_max(m);
+    float s = 0.f;
+    for (int64_t e = beg + lane; e < end; e += WAVE)
+      s += __expf(logits[e * H + h] - m);
+    s = wave_reduce_sum(s);
+    const float inv = 1.0f / fmaxf(s, 1e-38f);
+    for (int64_t e = beg + lane; e < end; e += WAVE)
+      alpha[e * H + h] = __expf(logits[e * H + h] - m) * inv;
+  }
+}
+
+// dlogit[e,h] = a[e,h]*(g[e,h] - sum_seg a*g)
+__global__ void segment_softmax_bwd_kernel(const int64_t* __restrict__ indptr,
+                                           const float* __restrict__ alpha,
+                                           const float* __restrict__ grad,
+                                           float* __restrict__ out,
+                                           int n_rows, int H) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int rh = wave; rh < n_rows * H; rh += n_waves) {
+    const int r = rh / H, h = rh % H;
+    const int64_t beg = indptr[r], end = indptr[r + 1];
+    float s = 0.f;
+    for (int64_t e = beg + lane; e < end; e += WAVE)
+      s += alpha[e * H + h] * grad[e * H + h];
+    s = wave_reduce_sum(s);
+    for (int64_t e = beg + lane; e < end; e += WAVE)
+      out[e * H + h] = alpha[e * H + h] * (grad[e * H + h] - s);
+  }
+}
+
+// out[r,h,:] (+)= sum_e w[e,h] * x[col_e,h,:] — wave per (row,head),
+// lanes stride D (coalesced within a head row).
+template <bool ACC>
+__global__ void spmm_edge_kernel(const int64_t* __restrict__ indptr,
+                                 const int32_t* __restrict__ indices,
+                                 const float* __restrict__ w,
+                                 const float* __restrict__ x,
+                                 float* __restrict__ out,
+                                 int n_rows, int H, int D) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int rh = wave; rh < n_rows * H; rh += n_waves) {
+    const int r = rh / H, h = rh % H;
+    const int64_t beg = indptr[r], end = indptr[r + 1];
+    for (int d = lane; d < D; d += WAVE) {
+      float acc = 0.f;
+      for (int64_t e = beg; e < end; ++e) {
+        const int c = indices[e];
+        acc += w[e * H + h] * x[((int64_t)c * H + h) * D + d];
+      }
+      const int64_t o = ((int64_t)r * H + h) * D + d;
+      if (ACC) out[o] += acc; else out[o] = acc;
+    }
+  }
+}
+
+// gw[e,h] = <g[row_e,h,:], x[col_e,h,:]> — wave per edge, lanes over D.
+__global__ void sddmm_dot_kernel(const int64_t* __restrict__ indptr,
+                                 const int32_t* __restrict__ indices,
+                                 const float* __restrict__ g,
+                                 const float* __restrict__ x,
+                                 float* __restrict__ out,
+                                 int n_rows, int64_t n_edges, int H, int D) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int r = wave; r < n_rows; r += n_waves) {
+    const int64_t beg = indptr[r], end = indptr[r + 1];
+    for (int64_t e = beg; e < end; ++e) {
+      const int c = indices[e];
+      for (int h = 0; h < H; ++h) {
+        float acc = 0.f;
+        for (int d = lane; d < D; d += WAVE)
+          acc += g[((int64_t)r * H + h) * D + d] *
+                 x[((int64_t)c * H + h) * D + d];
+        acc = wave_reduce_sum(acc);
+        if (lane == 0) out[e * H + h] = acc;
+      }
+    }
+  }
+}
+
+// ------------------------------ launchers ------------------------------
+
+inline void check_f32(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+                  t.scalar_type() == at::kFloat,
+              name, " must be contiguous fp32 CUDA tensor");
+}
+
+const float* opt_ptr(const c10::optional<at::Tensor>& t) {
+  return t.has_value() ? t->data_ptr<float>() : nullptr;
+}
+
+int spmm_grid(int n_rows) {
+  // wave-per-row, 4 rows per block; >> 256 workgroups fills 8 XCDs
+  const int blocks = (n_rows + 3) / 4;
+  return std::min(blocks, 16384);
+}
+
+at::Tensor spmm_sum(at::Tensor indptr, at::Tensor indices, at::Tensor x,
+                    c10::optional<at::Tensor> src_scale,
+                    c10::optional<at::Tensor> dst_scale,
+                    c10::optional<at::Tensor> out_opt) {
+  check_f32(x, "x");
+  TORCH_CHECK(indptr.scalar_type() == at::kLong && indices.scalar_type() == at::kInt,
+              "indptr int64 / indices int32 expected");
+  const int n_rows = indptr.numel() - 1;
+  const int F = x.size(1);
+  at::Tensor out;
+  const bool acc = out_opt.has_value();
+  if (acc) { out = *out_opt; check_f32(out, "out"); }
+  else out = at::empty({n_rows, F}, x.options());
+  if (n_rows == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int grid = spmm_grid(n_rows);
+  if (indices.numel() == 0 && !acc) { out.zero_(); }
+  if (F % 4 == 0) {
+    auto kfn = acc ? spmm_sum_vec4_kernel<true> : spmm_sum_vec4_kernel<false>;
+    hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
+                       indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
+                       x.data_ptr<float>(), opt_ptr(src_scale),
+                       opt_ptr(dst_scale), out.data_ptr<float>(), n_rows, F / 4);
+  } else {
+    auto kfn = acc ? spmm_sum_scalar_kernel<true> : spmm_sum_scalar_kernel<false>;
+    hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
+                       indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
+                       x.data_ptr<float>(), opt_ptr(src_scale),
+                       opt_ptr(dst_scale), out.data_ptr<float>(), n_rows, F);
+  }
+  return out;
+}
+
+at::Tensor pack_rows(at::Tensor x, at::Tensor idx,
+                     c10::optional<at::Tensor> scale) {
+  check_f32(x, "x");
+  TORCH_CHECK(idx.scalar_type() == at::kLong, "idx must be int64");
+  const int n = idx.numel(), F = x.size(1);
+  auto out = at::empty({n, F}, x.options());
+  if (n == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int64_t total = (int64_t)n * F;
+  const int grid = std::min<int64_t>((total + 255) / 256, 4096);
+  hipLaunchKernelGGL(pack_rows_kernel, dim3(grid), dim3(256), 0, stream,
+                     x.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                     opt_ptr(scale), out.data_ptr<float>(), n, F);
+  return out;
+}
+
+void scatter_add_rows(at::Tensor out, at::Tensor idx, at::Tensor src,
+                      c10::optional<at::Tensor> scale) {
+  check_f32(out, "out");
+  check_f32(src, "src");
+  const int n = idx.numel(), F = out.size(1);
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int64_t total = (int64_t)n * F;
+  const int grid = std::min<int64_t>((total + 255) / 256, 4096);
+  hipLaunchKernelGGL(scatter_add_rows_kernel, dim3(grid), dim3(256), 0, stream,
+                     out.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                     src.data_ptr<float>(), opt_ptr(scale), n, F);
+}
+
+at::Tensor philox_keys(int64_t n, int64_t seed, int64_t epoch,
+                       int64_t src_rank, int64_t dst_rank) {
+  auto out = at::empty({n}, at::dtype(at::kLong).device(at::kCUDA));
+  if (n == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const uint32_t sd = (((uint32_t)src_rank & 0xFFFF) << 16) |
+                      ((uint32_t)dst_rank & 0xFFFF);
+  hipLaunchKernelGGL(philox_keys_kernel, dim3((n + 255) / 256), dim3(256), 0,
+                     stream, out.data_ptr<int64_t>(), (int)n,
+                     (uint32_t)(seed & 0xFFFFFFFF),
+                     (uint32_t)((seed >> 32) & 0xFFFFFFFF),
+                     (uint32_t)(epoch & 0xFFFFFFFF), sd);
+  return out;
+}
+
+at::Tensor syncbn_stats(at::Tensor x) {
+  check_f32(x, "x");
+  const int64_t n = x.size(0);
+  const int F = x.size(1);
+  auto out = at::zeros({2, F}, x.options());
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int row_chunks = (int)std::min<int64_t>((n + 4095) / 4096, 64);
+  dim3 grid((F + 255) / 256, row_chunks);
+  hipLaunchKernelGGL(syncbn_stats_kernel, grid, dim3(256), 0, stream,
+                     x.data_ptr<float>(), out.data_ptr<float>(), n, F,
+                     row_chunks);
+  return out;
+}
+
+at::Tensor sddmm_add(at::Tensor indptr, at::Tensor indices, at::Tensor el,
+                     at::Tensor er) {
+  check_f32(el, "el"); check_f32(er, "er");
+  const int n_rows = indptr.numel() - 1;
+  const int H = el.size(1);
+  auto out = at::empty({indices.numel(), H}, el.options());
+  if (indices.numel() == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(sddmm_add_kernel, dim3(spmm_grid(n_rows)), dim3(256), 0,
+                     stream, indptr.data_ptr<int64_t>(),
+                     indices.data_ptr<int32_t>(), el.data_ptr<float>(),
+                     er.data_ptr<float>(), out.data_ptr<float>(), n_rows, H);
+  return out;
+}
+
+at::Tensor segment_softmax(at::Tensor indptr, at::Tensor logits) {
+  check_f32(logits, "logits");
+  const int n_rows = indptr.numel() - 1;
+  const int H = logits.size(1);
+  auto out = at::empty_like(logits);
+  if (logits.numel() == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(segment_softmax_kernel, dim3(spmm_grid(n_rows * H)),
+                     dim3(256), 0, stream, indptr.data_ptr<int64_t>(),
+                     logits.data_ptr<float>(), out.data_ptr<float>(), n_rows, H);
+  return out;
+}
+
+at::Tensor segment_softmax_backward(at::Tensor indptr, at::Tensor alpha,
+                                    at::Tensor grad) {
+  check_f32(alpha, "alpha"); check_f32(grad, "grad");
+  const int n_rows = indptr.numel() - 1;
+  const int H = alpha.size(1);
+  auto out = at::empty_like(alpha);
+  if (alpha.numel() == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(segment_softmax_bwd_kernel, dim3(spmm_grid(n_rows * H)),
+                     dim3(256), 0, stream, indptr.data_ptr<int64_t>(),
+                     alpha.data_ptr<float>(), grad.data_ptr<float>(),
+                     out.data_ptr<float>(), n_rows, H);
+  return out;
+}
+
+at::Tensor spmm_edge_sum(at::Tensor indptr, at::Tensor indices,
+                         at::Tensor w, at::Tensor x,
+                         c10::optional<at::Tensor> out_opt) {
+  check_f32(w, "w"); check_f32(x, "x");
+  const int n_rows = indptr.numel() - 1;
+  const int H = x.size(1), D = x.size(2);
+  at::Tensor out;
+  const bool acc = out_opt.has_value();
+  if (acc) out = *out_opt;
+  else out = at::zeros({n_rows, H, D}, x.options());
+  if (indices.numel() == 0 || n_rows == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  auto kfn = acc ? spmm_edge_kernel<true> : spmm_edge_kernel<false>;
+  if (!acc) {
+    kfn = spmm_edge_kernel<false>;
+  }
+  hipLaunchKernelGGL(kfn, dim3(spmm_grid(n_rows * H)), dim3(256), 0, stream,
+                     indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
+                     w.data_ptr<float>(), x.data_ptr<float>(),
+                     out.data_ptr<float>(), n_rows, H, D);
+  return out;
+}
+
+at::Tensor sddmm_dot(at::Tensor indptr, at::Tensor indices, at::Tensor g,
+                     at::Tensor x) {
+  check_f32(g, "g"); check_f32(x, "x");
+  const int n_rows = indptr.numel() - 1;
+  const int H = x.size(1), D = x.size(2);
+  auto out = at::empty({indices.numel(), H}, x.options());
+  if (indices.numel() == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(sddmm_dot_kernel, dim3(spmm_grid(n_rows)), dim3(256), 0,
+                     stream, indptr.data_ptr<int64_t>(),
+                     indices.data_ptr<int32_t>(), g.data_ptr<float>(),
+                     x.data_ptr<float>(), out.data_ptr<float>(), n_rows,
+                     indices.numel(), H, D);
+  return out;
+}
+
+at::Tensor gemm_strided(const at::Tensor& A, int64_t sAm, int64_t sAk,
+                        const at::Tensor& B, int64_t sBk, int64_t sBn,
+                        const c10::optional<at::Tensor>& bias,
+                        int M, int N, int K) {
+  auto C = at::empty({M, N}, A.options());
+  if (M == 0 || N == 0) return C;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+  hipLaunchKernelGGL(gemm_f32_kernel, grid, dim3(256), 0, stream,
+                     A.data_ptr<float>(), sAm, sAk, B.data_ptr<float>(), sBk,
+                     sBn, opt_ptr(bias), C.data_ptr<float>(), M, N, K);
+  return C;
+}
+
+// y[M,N] = x[M,K] @ w[N,K]^T + b
+at::Tensor gemm_nt_bias(at::Tensor x, at::Tensor w,
+                        c10::optional<at::Tensor> bias) {
+  check_f32(x, "x"); check_f32(w, "w");
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "shape mismatch");
+  return gemm_strided(x, K, 1, w, 1, K, bias, M, N, K);
+}
+
+// dx[M,K] = g[M,N] @ w[N,K]
+at::Tensor gemm_nn(at::Tensor g, at::Tensor w) {
+  check_f32(g, "g"); check_f32(w, "w");
+  const int M = g.size(0), N = g.size(1), K = w.size(1);
+  return gemm_strided(g, N, 1, w, K, 1, c10::nullopt, M, K, N);
+}
+
+// dw[N,K] = g[M,N]^T @ x[M,K]
+at::Tensor gemm_tn(at::Tensor g, at::Tensor x) {
+  check_f32(g, "g"); check_f32(x, "x");
+  const int M = g.size(0), N = g.size(1), K = x.size(1);
+  return gemm_strided(g, 1, N, x, K, 1, c10::nullopt, N, K, M);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("spmm_sum", &spmm_sum, "CSR SpMM sum with fused src/dst scales");
+  m.def("pack_rows", &pack_rows, "gather rows + per-row scale");
+  m.def("scatter_add_rows", &scatter_add_rows, "scatter-add rows + scale");
+  m.def("philox_keys", &philox_keys, "BNS sampling keys (Philox4x32-10)");
+  m.def("syncbn_stats", &syncbn_stats, "column sum + sum of squares");
+  m.def("gemm_nt_bias", &gemm_nt_bias, "fp32 MFMA GEMM x@w^T+b");
+  m.def("gemm_nn", &gemm_nn, "fp32 MFMA GEMM g@w");
+  m.def("gemm_tn", &gemm_tn, "fp32 MFMA GEMM g^T@x");
+  m.def("sddmm_add", &sddmm_add, "GAT u_add_v SDDMM");
+  m.def("segment_softmax", &segment_softmax, "edge softmax by dst segment");
+  m.def("segment_softmax_backward", &segment_softmax_backward,
+        "edge softmax backward");
+  m.def("spmm_edge_sum", &spmm_edge_sum, "multi-head edge-weighted SpMM");
+  m.def("sddmm_dot", &sddmm_dot, "per-edge per-head dot (spmm_edge grad)");
+}

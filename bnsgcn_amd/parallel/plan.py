@@ -84,6 +84,23 @@ class HaloPlan:
         self._state: EpochState | None = None
         self._static = self.rate >= 1.0 or self.rate <= 0.0
 
+    def _sample(self, n: int, s: int, epoch: int, src: int, dst: int) -> torch.Tensor:
+        """Sorted s-subset of [0,n) — device-side Philox + torch stable sort
+        on GPU (reference K16 was CPU np.random.choice),
+        numpy Philox on CPU; both bitwise-identical."""
+        dev = self.device
+        if s >= n:
+            return torch.arange(n, dtype=torch.long, device=dev)
+        if dev.type == "cuda":
+            from ..ops._ext import get_ext, has_ext
+            if has_ext():
+                with torch.cuda.device(dev):
+                    keys = get_ext().philox_keys(n, self.seed, epoch, src, dst)
+                order = torch.argsort(keys, stable=True)[:s]
+                return torch.sort(order)[0]
+        return torch.from_numpy(
+            sample_boundary(n, s, self.seed, epoch, src, dst)).to(dev)
+
     # ------------------------------------------------------------------
     def set_epoch(self, epoch: int) -> EpochState:
         if self._static and self._state is not None:
@@ -97,17 +114,15 @@ class HaloPlan:
             # outgoing sample (me -> j)
             s, n = self.send_size[j], self.n_out[j]
             if s > 0:
-                pos = torch.from_numpy(
-                    sample_boundary(n, s, self.seed, epoch, me, j)).to(dev)
+                pos = self._sample(n, s, epoch, me, j)
                 pack_parts.append(self.boundary[j][pos])
                 if not self.unit_ratio:
                     scale_parts.append(torch.full((s,), n / s, dtype=torch.float32,
-                                                  device=dev))
+                                                  device=self.device))
             # incoming sample (j -> me)
             r, m = self.recv_size[j], self.n_in[j]
             if r > 0:
-                pos = torch.from_numpy(
-                    sample_boundary(m, r, self.seed, epoch, j, me)).to(dev)
+                pos = self._sample(m, r, epoch, j, me)
                 hsel_parts.append(self.halo_start[j] + pos)
 
         pack_idx = (torch.cat(pack_parts) if pack_parts
