@@ -57,6 +57,11 @@ def all_to_all_rows(recv: torch.Tensor, send: torch.Tensor,
     order; recv: preallocated [sum(recv_counts), F]. Returns a waitable
     handle when async_op (nccl) else None.
     """
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        # single-process: only the (normally empty) self block
+        if recv_counts and recv_counts[0] > 0:
+            recv[:recv_counts[0]].copy_(send[:send_counts[0]])
+        return None
     if _supports_alltoall():
         return dist.all_to_all_single(recv, send,
                                       output_split_sizes=recv_counts,
@@ -91,6 +96,8 @@ def exchange_counts(my_counts: torch.Tensor) -> torch.Tensor:
     """Each rank contributes a length-P int64 vector (what I send to each
     peer); returns the length-P vector of what each peer sends to ME.
     Implemented with all_gather so it works on every backend."""
+    if not dist.is_initialized():
+        return my_counts.clone()
     size = dist.get_world_size()
     rank = dist.get_rank()
     gathered = [torch.zeros_like(my_counts) for _ in range(size)]
