@@ -46,6 +46,12 @@ class GraphContext:
         self.out_norm_inv = _inv_sqrt(out_deg)     # GCN src scale
         self.in_deg_inv = _inv(in_deg)             # SAGE mean dst scale
         self.n_rows = self.indptr.numel() - 1
+        if dev.type == "cuda":
+            # pre-build SpMM work lists on the default stream (avoids
+            # cross-stream allocation of cached plan tensors)
+            from ..ops.functional import _worklist_of
+            _worklist_of(self.indptr)
+            _worklist_of(self.t_indptr)
         # lazily built per-epoch GAT block (combined inner+halo CSR)
         self._gat_cache: tuple[int, tuple] | None = None
         # full-halo (p=1.0) exchange state for precompute / GAT layer 0

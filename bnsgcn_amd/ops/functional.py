@@ -17,9 +17,22 @@ from ._ext import get_ext, use_hip
 
 # ---------------------------------------------------------------- raw ops
 
+def _worklist_of(indptr):
+    """Edge-balanced work list for the device SpMM (csr_torch.build_worklist),
+    cached on the indptr tensor itself (same lifetime as the CSR)."""
+    wl = getattr(indptr, "_bns_worklist", None)
+    if wl is None:
+        from .csr_torch import build_worklist
+        wl = build_worklist(indptr)
+        indptr._bns_worklist = wl
+    return wl
+
+
 def spmm_sum_raw(indptr, indices, x, src_scale=None, dst_scale=None, out=None):
     if use_hip(x):
-        return get_ext().spmm_sum(indptr, indices, x, src_scale, dst_scale, out)
+        wrow, wbeg, wend = _worklist_of(indptr)
+        return get_ext().spmm_sum(wrow, wbeg, wend, indices, x,
+                                  indptr.numel() - 1, src_scale, dst_scale, out)
     return ref.spmm_sum(indptr, indices, x, src_scale, dst_scale, out)
 
 

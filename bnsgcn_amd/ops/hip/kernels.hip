@@ -15,71 +15,139 @@
 namespace {
 
 // ============================== SpMM ===================================
-// out[r, :] (+)= dst_scale[r] * sum_{e in row r} src_scale[col_e] * x[col_e, :]
-// One wave per output row (4 rows / 256-thread block), grid-stride over
-// rows. Feature dim covered by the 64 lanes: float4 path when F % 4 == 0
-// (64 lanes x 16 B = 1 KiB per neighbor row read — full coalescing),
-// scalar path otherwise. Rows are independent -> no atomics; the backward
-// pass reuses this kernel on the precomputed transposed CSR.
+// out[row, :] (+)= dst_scale[row] * sum_{e} src_scale[col_e] * x[col_e, :]
+// Work-list driven: each item = (row, edge range<=SEG) built by
+// ops/csr_torch.build_worklist — heavy (power-law hub) rows are split
+// across items and combined with atomicAdd, so no wave serializes a
+// 100k+-edge row. Per 64-edge chunk the wave loads 64 column ids (and
+// src scales) with ONE coalesced load each and broadcasts them via
+// __shfl, so the row-feature loads are address-independent and the
+// unrolled inner loop keeps several 1-KiB row reads in flight (the naive
+// row-loop form serializes two dependent loads per edge behind
+// s_waitcnt vmcnt(0) — measured 25x slower).
+// Feature dim is covered in passes of 2 float4 per lane (512 floats);
+// scalar variant (4 floats/lane/pass) handles F % 4 != 0.
+
+DEV_INLINE void f4_axpy(float4& a, float ss, const float4 v) {
+  a.x += ss * v.x; a.y += ss * v.y; a.z += ss * v.z; a.w += ss * v.w;
+}
 
 template <bool ACC>
-__global__ void spmm_sum_vec4_kernel(
-    const int64_t* __restrict__ indptr, const int32_t* __restrict__ indices,
-    const float* __restrict__ x, const float* __restrict__ src_scale,
-    const float* __restrict__ dst_scale, float* __restrict__ out,
-    int n_rows, int f4 /* F/4 */) {
+__global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
+    const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
+    const int64_t* __restrict__ wend, int n_items,
+    const int32_t* __restrict__ indices, const float* __restrict__ x,
+    const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
+    float* __restrict__ out, int f4) {
   const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n_waves = (gridDim.x * blockDim.x) / WAVE;
   const float4* __restrict__ x4 = reinterpret_cast<const float4*>(x);
   float4* __restrict__ out4 = reinterpret_cast<float4*>(out);
 
-  for (int r = wave; r < n_rows; r += n_waves) {
-    const int64_t beg = indptr[r], end = indptr[r + 1];
-    const float ds = dst_scale ? dst_scale[r] : 1.0f;
-    for (int f = lane; f < f4; f += WAVE) {
-      float4 acc = {0.f, 0.f, 0.f, 0.f};
-      for (int64_t e = beg; e < end; ++e) {
-        const int c = indices[e];
-        const float s = src_scale ? src_scale[c] : 1.0f;
-        const float4 v = x4[(int64_t)c * f4 + f];
-        acc.x += s * v.x; acc.y += s * v.y;
-        acc.z += s * v.z; acc.w += s * v.w;
+  for (int it = wave; it < n_items; it += n_waves) {
+    int row = wrow[it];
+    const bool atomic = row < 0;
+    if (atomic) row = ~row;
+    const int64_t beg = wbeg[it], end = wend[it];
+    const float ds = dst_scale ? dst_scale[row] : 1.0f;
+    for (int f0 = 0; f0 < f4; f0 += 2 * WAVE) {
+      const int fA = f0 + lane;
+      const int fB = fA + WAVE;
+      const bool hasA = fA < f4, hasB = fB < f4;
+      float4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+      for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
+        const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
+        int cid = 0;
+        float ssc = 1.0f;
+        if (lane < nv) {
+          cid = indices[e0 + lane];
+          if (src_scale) ssc = src_scale[cid];
+        }
+#pragma unroll 4
+        for (int k = 0; k < nv; ++k) {
+          const int c = __shfl(cid, k, WAVE);
+          const float ss = src_scale ? __shfl(ssc, k, WAVE) : 1.0f;
+          const int64_t base = (int64_t)c * f4;
+          if (hasA) f4_axpy(acc0, ss, x4[base + fA]);
+          if (hasB) f4_axpy(acc1, ss, x4[base + fB]);
+        }
       }
-      const int64_t o = (int64_t)r * f4 + f;
-      if (ACC) {
-        float4 prev = out4[o];
-        prev.x += ds * acc.x; prev.y += ds * acc.y;
-        prev.z += ds * acc.z; prev.w += ds * acc.w;
-        out4[o] = prev;
+      const int64_t ob = (int64_t)row * f4;
+      if (atomic) {
+        if (hasA) {
+          float* p = reinterpret_cast<float*>(&out4[ob + fA]);
+          atomicAdd(p + 0, ds * acc0.x); atomicAdd(p + 1, ds * acc0.y);
+          atomicAdd(p + 2, ds * acc0.z); atomicAdd(p + 3, ds * acc0.w);
+        }
+        if (hasB) {
+          float* p = reinterpret_cast<float*>(&out4[ob + fB]);
+          atomicAdd(p + 0, ds * acc1.x); atomicAdd(p + 1, ds * acc1.y);
+          atomicAdd(p + 2, ds * acc1.z); atomicAdd(p + 3, ds * acc1.w);
+        }
       } else {
-        out4[o] = make_float4(ds * acc.x, ds * acc.y, ds * acc.z, ds * acc.w);
+        if (hasA) {
+          float4 v = make_float4(ds * acc0.x, ds * acc0.y, ds * acc0.z, ds * acc0.w);
+          if (ACC) { float4 pv = out4[ob + fA]; v.x += pv.x; v.y += pv.y; v.z += pv.z; v.w += pv.w; }
+          out4[ob + fA] = v;
+        }
+        if (hasB) {
+          float4 v = make_float4(ds * acc1.x, ds * acc1.y, ds * acc1.z, ds * acc1.w);
+          if (ACC) { float4 pv = out4[ob + fB]; v.x += pv.x; v.y += pv.y; v.z += pv.z; v.w += pv.w; }
+          out4[ob + fB] = v;
+        }
       }
     }
   }
 }
 
 template <bool ACC>
-__global__ void spmm_sum_scalar_kernel(
-    const int64_t* __restrict__ indptr, const int32_t* __restrict__ indices,
-    const float* __restrict__ x, const float* __restrict__ src_scale,
-    const float* __restrict__ dst_scale, float* __restrict__ out,
-    int n_rows, int F) {
+__global__ __launch_bounds__(256) void spmm_sum_scalar_kernel(
+    const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
+    const int64_t* __restrict__ wend, int n_items,
+    const int32_t* __restrict__ indices, const float* __restrict__ x,
+    const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
+    float* __restrict__ out, int F) {
   const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n_waves = (gridDim.x * blockDim.x) / WAVE;
-  for (int r = wave; r < n_rows; r += n_waves) {
-    const int64_t beg = indptr[r], end = indptr[r + 1];
-    const float ds = dst_scale ? dst_scale[r] : 1.0f;
-    for (int f = lane; f < F; f += WAVE) {
-      float acc = 0.f;
-      for (int64_t e = beg; e < end; ++e) {
-        const int c = indices[e];
-        const float s = src_scale ? src_scale[c] : 1.0f;
-        acc += s * x[(int64_t)c * F + f];
+  for (int it = wave; it < n_items; it += n_waves) {
+    int row = wrow[it];
+    const bool atomic = row < 0;
+    if (atomic) row = ~row;
+    const int64_t beg = wbeg[it], end = wend[it];
+    const float ds = dst_scale ? dst_scale[row] : 1.0f;
+    for (int f0 = 0; f0 < F; f0 += 4 * WAVE) {
+      float acc[4] = {0.f, 0.f, 0.f, 0.f};
+      for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
+        const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
+        int cid = 0;
+        float ssc = 1.0f;
+        if (lane < nv) {
+          cid = indices[e0 + lane];
+          if (src_scale) ssc = src_scale[cid];
+        }
+#pragma unroll 2
+        for (int k = 0; k < nv; ++k) {
+          const int c = __shfl(cid, k, WAVE);
+          const float ss = src_scale ? __shfl(ssc, k, WAVE) : 1.0f;
+          const int64_t base = (int64_t)c * F;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const int f = f0 + j * WAVE + lane;
+            if (f < F) acc[j] += ss * x[base + f];
+          }
+        }
       }
-      const int64_t o = (int64_t)r * F + f;
-      if (ACC) out[o] += ds * acc; else out[o] = ds * acc;
+      const int64_t ob = (int64_t)row * F;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int f = f0 + j * WAVE + lane;
+        if (f >= F) break;
+        if (atomic) atomicAdd(&out[ob + f], ds * acc[j]);
+        else if (ACC) out[ob + f] += ds * acc[j];
+        else out[ob + f] = ds * acc[j];
+      }
     }
   }
 }
@@ -187,15 +255,20 @@ void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
   f32x4 acc[2][2] = {};
 
   for (int k0 = 0; k0 < K; k0 += BK) {
-    // cooperative staging: 256 threads, A: 64x16 (4 elems/thread),
-    // B: 16x64 (4 elems/thread); bounds-checked (zero-fill).
+    // cooperative staging: 256 threads, 4 elems each, bounds-checked
+    // (zero-fill). The thread->element map follows the operand's unit
+    // stride so global reads stay coalesced for every trans layout.
     for (int t = tid; t < BM * BK; t += 256) {
-      const int i = t / BK, k = t % BK;
+      int i, k;
+      if (sAk == 1) { i = t / BK; k = t % BK; }   // row-major A
+      else          { i = t % BM; k = t / BM; }   // col-major A
       const int gm = m0 + i, gk = k0 + k;
       As[i][k] = (gm < M && gk < K) ? A[gm * sAm + gk * sAk] : 0.f;
     }
     for (int t = tid; t < BK * BN; t += 256) {
-      const int k = t / BN, j = t % BN;
+      int k, j;
+      if (sBn == 1) { k = t / BN; j = t % BN; }   // row-major B
+      else          { k = t % BK; j = t / BK; }   // col-major B
       const int gk = k0 + k, gn = n0 + j;
       Bs[k][j] = (gk < K && gn < N) ? B[gk * sBk + gn * sBn] : 0.f;
     }
@@ -392,35 +465,41 @@ int spmm_grid(int n_rows) {
   return std::min(blocks, 16384);
 }
 
-at::Tensor spmm_sum(at::Tensor indptr, at::Tensor indices, at::Tensor x,
+at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
+                    at::Tensor indices, at::Tensor x, int64_t n_rows,
                     c10::optional<at::Tensor> src_scale,
                     c10::optional<at::Tensor> dst_scale,
                     c10::optional<at::Tensor> out_opt) {
   check_f32(x, "x");
-  TORCH_CHECK(indptr.scalar_type() == at::kLong && indices.scalar_type() == at::kInt,
-              "indptr int64 / indices int32 expected");
-  const int n_rows = indptr.numel() - 1;
+  TORCH_CHECK(wrow.scalar_type() == at::kInt &&
+                  wbeg.scalar_type() == at::kLong &&
+                  indices.scalar_type() == at::kInt,
+              "worklist int32/int64 + indices int32 expected");
   const int F = x.size(1);
+  const int n_items = wrow.numel();
   at::Tensor out;
   const bool acc = out_opt.has_value();
   if (acc) { out = *out_opt; check_f32(out, "out"); }
-  else out = at::empty({n_rows, F}, x.options());
-  if (n_rows == 0) return out;
+  else { out = at::zeros({n_rows, F}, x.options()); }
+  if (n_items == 0) return out;
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int grid = spmm_grid(n_rows);
-  if (indices.numel() == 0 && !acc) { out.zero_(); }
+  const int grid = spmm_grid(n_items);
   if (F % 4 == 0) {
     auto kfn = acc ? spmm_sum_vec4_kernel<true> : spmm_sum_vec4_kernel<false>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
-                       indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                       x.data_ptr<float>(), opt_ptr(src_scale),
-                       opt_ptr(dst_scale), out.data_ptr<float>(), n_rows, F / 4);
+                       wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
+                       wend.data_ptr<int64_t>(), n_items,
+                       indices.data_ptr<int32_t>(), x.data_ptr<float>(),
+                       opt_ptr(src_scale), opt_ptr(dst_scale),
+                       out.data_ptr<float>(), F / 4);
   } else {
     auto kfn = acc ? spmm_sum_scalar_kernel<true> : spmm_sum_scalar_kernel<false>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
-                       indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                       x.data_ptr<float>(), opt_ptr(src_scale),
-                       opt_ptr(dst_scale), out.data_ptr<float>(), n_rows, F);
+                       wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
+                       wend.data_ptr<int64_t>(), n_items,
+                       indices.data_ptr<int32_t>(), x.data_ptr<float>(),
+                       opt_ptr(src_scale), opt_ptr(dst_scale),
+                       out.data_ptr<float>(), F);
   }
   return out;
 }

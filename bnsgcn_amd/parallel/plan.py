@@ -135,6 +135,10 @@ class HaloPlan:
 
         bwd_ip, bwd_ix = gather_rows_csr(self.halo_indptr, self.halo_indices, hsel)
         fwd_ip, fwd_ix, _ = transpose_csr(bwd_ip, bwd_ix, self.n_inner)
+        if dev.type == "cuda":
+            from ..ops.functional import _worklist_of
+            _worklist_of(bwd_ip)
+            _worklist_of(fwd_ip)
 
         st = EpochState(
             epoch=epoch,

@@ -31,23 +31,46 @@ def rand_csr(n_rows, n_cols, e, seed=0):
 
 
 @needs_gpu
-@pytest.mark.parametrize("F", [256, 602, 17])
+@pytest.mark.parametrize("F", [256, 602, 17, 512])
 def test_spmm_matches_reference(F):
+    from bnsgcn_amd.ops.functional import spmm_sum_raw
     torch.manual_seed(0)
     indptr, indices = rand_csr(500, 700, 6000)
     x = torch.randn(700, F)
     ss = torch.rand(700) + 0.5
     ds = torch.rand(500) + 0.5
     want = ref.spmm_sum(indptr, indices, x, ss, ds)
-    got = ext.spmm_sum(indptr.cuda(), indices.cuda(), x.cuda(), ss.cuda(),
-                       ds.cuda(), None).cpu()
+    got = spmm_sum_raw(indptr.cuda(), indices.cuda(), x.cuda(), ss.cuda(),
+                       ds.cuda()).cpu()
     torch.testing.assert_close(got, want, rtol=2e-5, atol=1e-5)
     # accumulate path + no-scale path
     base = torch.randn(500, F)
-    got2 = ext.spmm_sum(indptr.cuda(), indices.cuda(), x.cuda(), None, None,
+    got2 = spmm_sum_raw(indptr.cuda(), indices.cuda(), x.cuda(), None, None,
                         base.clone().cuda()).cpu()
     want2 = ref.spmm_sum(indptr, indices, x, None, None, base.clone())
     torch.testing.assert_close(got2, want2, rtol=2e-5, atol=1e-5)
+
+
+@needs_gpu
+def test_spmm_heavy_row_split():
+    """Power-law CSR with a hub row >> SEG: exercises the work-list split
+    + atomic combine path."""
+    from bnsgcn_amd.ops.functional import spmm_sum_raw
+    rng = np.random.default_rng(3)
+    n_rows, n_cols = 300, 400
+    dst = np.concatenate([np.zeros(9000, dtype=np.int64),          # hub row
+                          rng.integers(0, n_rows, 4000)])
+    src = rng.integers(0, n_cols, dst.shape[0])
+    c = CSR.from_edges(src, dst, n_rows, n_cols)
+    indptr, indices = torch.from_numpy(c.indptr), torch.from_numpy(c.indices)
+    for F in (256, 31):
+        x = torch.randn(n_cols, F)
+        ss = torch.rand(n_cols) + 0.5
+        ds = torch.rand(n_rows) + 0.5
+        want = ref.spmm_sum(indptr, indices, x, ss, ds)
+        got = spmm_sum_raw(indptr.cuda(), indices.cuda(), x.cuda(), ss.cuda(),
+                           ds.cuda()).cpu()
+        torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-3)
 
 
 @needs_gpu

@@ -224,3 +224,26 @@ def test_linear_matches_torch():
     torch.testing.assert_close(x.grad, x2.grad)
     torch.testing.assert_close(w.grad, w2.grad)
     torch.testing.assert_close(b.grad, b2.grad)
+
+
+def test_build_worklist():
+    from bnsgcn_amd.ops.csr_torch import build_worklist
+    # degrees: 0, 3, 5000, 10 with seg=2048 -> row 2 split into 3 items
+    indptr = torch.tensor([0, 0, 3, 5003, 5013], dtype=torch.int64)
+    wrow, wbeg, wend = build_worklist(indptr, seg=2048)
+    assert wrow.numel() == 1 + 1 + 3 + 1
+    # every edge covered exactly once, rows correct
+    cover = torch.zeros(5013, dtype=torch.int32)
+    for r, b, e in zip(wrow.tolist(), wbeg.tolist(), wend.tolist()):
+        row = ~r if r < 0 else r
+        lo, hi = int(indptr[row]), int(indptr[row + 1])
+        assert lo <= b <= e <= hi
+        cover[b:e] += 1
+    assert (cover == 1).all()
+    # split flag only on the heavy row
+    for r in wrow.tolist():
+        if r < 0:
+            assert ~r == 2
+    # longest-first
+    lens = (wend - wbeg).tolist()
+    assert lens == sorted(lens, reverse=True)
