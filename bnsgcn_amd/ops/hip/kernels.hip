@@ -237,11 +237,15 @@ __global__ void syncbn_stats_kernel(const float* __restrict__ x,
 #define BN 64
 #define BK 16
 
+// ATOMIC variant: split-K — blockIdx.z picks a K slice, partials combine
+// via atomicAdd (used when M/N tiles alone cannot fill 256 CUs, e.g. the
+// dW = g^T @ x reduction GEMMs with K = node count).
+template <bool ATOMIC>
 __global__ __launch_bounds__(256)
 void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
                      const float* __restrict__ B, int64_t sBk, int64_t sBn,
                      const float* __restrict__ bias, float* __restrict__ C,
-                     int M, int N, int K) {
+                     int M, int N, int K, int k_slice) {
   __shared__ float As[BM][BK + 1];
   __shared__ float Bs[BK][BN + 1];
   const int m0 = blockIdx.x * BM;
@@ -254,7 +258,9 @@ void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
   using f32x4 = __attribute__((ext_vector_type(4))) float;
   f32x4 acc[2][2] = {};
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
+  const int kb = blockIdx.z * k_slice;
+  const int ke = (kb + k_slice < K) ? kb + k_slice : K;
+  for (int k0 = kb; k0 < ke; k0 += BK) {
     // cooperative staging: 256 threads, 4 elems each, bounds-checked
     // (zero-fill). The thread->element map follows the operand's unit
     // stride so global reads stay coalesced for every trans layout.
@@ -263,14 +269,14 @@ void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
       if (sAk == 1) { i = t / BK; k = t % BK; }   // row-major A
       else          { i = t % BM; k = t / BM; }   // col-major A
       const int gm = m0 + i, gk = k0 + k;
-      As[i][k] = (gm < M && gk < K) ? A[gm * sAm + gk * sAk] : 0.f;
+      As[i][k] = (gm < M && gk < ke) ? A[gm * sAm + gk * sAk] : 0.f;
     }
     for (int t = tid; t < BK * BN; t += 256) {
       int k, j;
       if (sBn == 1) { k = t / BN; j = t % BN; }   // row-major B
       else          { k = t % BK; j = t / BK; }   // col-major B
       const int gk = k0 + k, gn = n0 + j;
-      Bs[k][j] = (gk < K && gn < N) ? B[gk * sBk + gn * sBn] : 0.f;
+      Bs[k][j] = (gk < ke && gn < N) ? B[gk * sBk + gn * sBn] : 0.f;
     }
     __syncthreads();
 #pragma unroll
@@ -303,7 +309,8 @@ void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
         if (gm < M && gn < N) {
           float v = acc[fi][fj][r];
           if (bias) v += bias[gn];
-          C[(int64_t)gm * N + gn] = v;
+          if (ATOMIC) atomicAdd(&C[(int64_t)gm * N + gn], v);
+          else C[(int64_t)gm * N + gn] = v;
         }
       }
 }
@@ -649,13 +656,31 @@ at::Tensor gemm_strided(const at::Tensor& A, int64_t sAm, int64_t sAk,
                         const at::Tensor& B, int64_t sBk, int64_t sBn,
                         const c10::optional<at::Tensor>& bias,
                         int M, int N, int K) {
-  auto C = at::empty({M, N}, A.options());
-  if (M == 0 || N == 0) return C;
   auto stream = at::cuda::getCurrentCUDAStream();
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
-  hipLaunchKernelGGL(gemm_f32_kernel, grid, dim3(256), 0, stream,
+  // split-K when the M/N tile grid alone cannot fill the 256 CUs (dW
+  // reductions: C is tiny, K is the node count)
+  const int base_blocks = grid.x * grid.y;
+  int splitk = 1;
+  if (base_blocks < 256 && K > 4 * BK) {
+    splitk = std::min({(int)(768 / base_blocks), (K + 255) / 256, 64});
+    splitk = std::max(splitk, 1);
+  }
+  if (splitk > 1) {
+    TORCH_CHECK(!bias.has_value(), "split-K GEMM does not take a bias");
+    auto C = at::zeros({M, N}, A.options());
+    const int k_slice = ((K + splitk - 1) / splitk + BK - 1) / BK * BK;
+    grid.z = (K + k_slice - 1) / k_slice;
+    hipLaunchKernelGGL(gemm_f32_kernel<true>, grid, dim3(256), 0, stream,
+                       A.data_ptr<float>(), sAm, sAk, B.data_ptr<float>(), sBk,
+                       sBn, nullptr, C.data_ptr<float>(), M, N, K, k_slice);
+    return C;
+  }
+  auto C = at::empty({M, N}, A.options());
+  if (M == 0 || N == 0) return C;
+  hipLaunchKernelGGL(gemm_f32_kernel<false>, grid, dim3(256), 0, stream,
                      A.data_ptr<float>(), sAm, sAk, B.data_ptr<float>(), sBk,
-                     sBn, opt_ptr(bias), C.data_ptr<float>(), M, N, K);
+                     sBn, opt_ptr(bias), C.data_ptr<float>(), M, N, K, K);
   return C;
 }
 
