@@ -34,41 +34,52 @@ def transpose_csr(indptr: torch.Tensor, indices: torch.Tensor, n_cols: int
 SEG = 2048  # max edges per SpMM work item (heavy-row split granularity)
 
 
-def build_worklist(indptr: torch.Tensor, seg: int = SEG
-                   ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-    """Edge-balanced work list for the gfx950 SpMM kernel.
+def build_worklist(indptr: torch.Tensor, seg: int = SEG, max_waves: int = 65536
+                   ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Edge-balanced work schedule for the gfx950 SpMM kernel.
 
-    Graph degree distributions are power-law (synthetic Reddit max degree
-    ~750k vs mean ~490): a plain wave-per-row mapping leaves one wave
-    grinding the hub row for most of the kernel. Each work item covers at
-    most `seg` edges of one row; split rows combine via atomicAdd (their
-    item row id is bitwise-negated). Items are sorted longest-first so the
-    scheduler starts the big ones early.
+    Three ideas (each measured on the Reddit-shaped bench):
+    * heavy-row split: power-law hubs (max degree ~750k vs mean ~490) are
+      split into <= `seg`-edge items combined via atomicAdd (item row id
+      bitwise-negated), so no single wave serializes a hub row;
+    * items stay in natural (ascending dst) order: the locality partitioner
+      gives consecutive rows overlapping neighborhoods, so consecutive
+      waves re-hit the same feature rows in L2;
+    * per-wave contiguous ranges balanced by EDGE count (wave_start), with
+      the kernel's XCD-aware block remap giving each XCD chiplet a
+      contiguous slab of the row space for its private L2.
 
-    Returns (wrow int32 [W] (negative ~row = atomic), wbeg int64, wend
-    int64), device-resident alongside indptr.
+    Returns (wrow int32 [W_items] (negative ~row = atomic), wbeg, wend
+    int64, wave_start int32 [n_waves+1]); n_waves is a multiple of 4
+    (one 256-thread block = 4 waves).
     """
     device = indptr.device
     deg = (indptr[1:] - indptr[:-1])
     n = deg.numel()
-    nseg = (deg + (seg - 1)) // seg
-    nseg = torch.clamp(nseg, min=1)
+    nseg = torch.clamp((deg + (seg - 1)) // seg, min=1)
     total = int(nseg.sum())
     rows = torch.repeat_interleave(torch.arange(n, device=device), nseg)
-    # offset of each item within its row
-    item_first = torch.zeros(n, dtype=torch.long, device=device)
-    torch.cumsum(nseg, 0, out=item_first[0:])  # exclusive via shift below
-    item_first = torch.cat([torch.zeros(1, dtype=torch.long, device=device),
+    item_first = torch.cumsum(nseg, 0)
+    item_first = torch.cat([torch.zeros(1, dtype=item_first.dtype, device=device),
                             item_first[:-1]])
     k = torch.arange(total, device=device) - item_first[rows]
     wbeg = indptr[rows] + k * seg
     wend = torch.minimum(wbeg + seg, indptr[rows + 1])
     split = nseg[rows] > 1
     wrow = torch.where(split, ~rows, rows).to(torch.int32)
-    # longest-first order
-    order = torch.argsort(wbeg - wend)  # ascending (beg-end) = descending len
-    return wrow[order].contiguous(), wbeg[order].contiguous(), \
-        wend[order].contiguous()
+
+    lens = (wend - wbeg)
+    cum = torch.cumsum(torch.cat([torch.zeros(1, dtype=lens.dtype, device=device),
+                                  lens]), 0)
+    n_waves = min(max_waves, max(4, total))
+    n_waves = (n_waves + 3) // 4 * 4
+    targets = (cum[-1] * torch.arange(n_waves + 1, device=device).double()
+               / n_waves).to(cum.dtype)
+    wave_start = torch.searchsorted(cum, targets).to(torch.int32)
+    wave_start[0] = 0
+    wave_start[-1] = total
+    return (wrow.contiguous(), wbeg.contiguous(), wend.contiguous(),
+            wave_start.contiguous())
 
 
 def merge_csr(ip1: torch.Tensor, ix1: torch.Tensor,

@@ -30,8 +30,8 @@ def _worklist_of(indptr):
 
 def spmm_sum_raw(indptr, indices, x, src_scale=None, dst_scale=None, out=None):
     if use_hip(x):
-        wrow, wbeg, wend = _worklist_of(indptr)
-        return get_ext().spmm_sum(wrow, wbeg, wend, indices, x,
+        wrow, wbeg, wend, wave_start = _worklist_of(indptr)
+        return get_ext().spmm_sum(wrow, wbeg, wend, wave_start, indices, x,
                                   indptr.numel() - 1, src_scale, dst_scale, out)
     return ref.spmm_sum(indptr, indices, x, src_scale, dst_scale, out)
 

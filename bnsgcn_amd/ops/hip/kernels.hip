@@ -32,20 +32,31 @@ DEV_INLINE void f4_axpy(float4& a, float ss, const float4 v) {
   a.x += ss * v.x; a.y += ss * v.y; a.z += ss * v.z; a.w += ss * v.w;
 }
 
+// XCD-aware bijective block remap (dispatcher places block b on XCD b%8):
+// XCD x gets a CONTIGUOUS range of logical blocks, so consecutive waves on
+// one chiplet touch consecutive worklist items -> dst-row locality in the
+// XCD-private L2 (cdna_hip_programming.md T1; bijective form for nb%8!=0).
+DEV_INLINE int xcd_remap_block(int b, int nb) {
+  const int q = nb / 8, r = nb % 8;
+  const int xcd = b % 8, j = b / 8;
+  return (xcd < r) ? xcd * (q + 1) + j : r * (q + 1) + (xcd - r) * q + j;
+}
+
 template <bool ACC>
 __global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
     const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
-    const int64_t* __restrict__ wend, int n_items,
+    const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
     const int32_t* __restrict__ indices, const float* __restrict__ x,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
     float* __restrict__ out, int f4) {
-  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
+  const int w = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
-  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const int it_beg = wave_start[w], it_end = wave_start[w + 1];
   const float4* __restrict__ x4 = reinterpret_cast<const float4*>(x);
   float4* __restrict__ out4 = reinterpret_cast<float4*>(out);
 
-  for (int it = wave; it < n_items; it += n_waves) {
+  for (int it = it_beg; it < it_end; ++it) {
     int row = wrow[it];
     const bool atomic = row < 0;
     if (atomic) row = ~row;
@@ -104,14 +115,15 @@ __global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
 template <bool ACC>
 __global__ __launch_bounds__(256) void spmm_sum_scalar_kernel(
     const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
-    const int64_t* __restrict__ wend, int n_items,
+    const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
     const int32_t* __restrict__ indices, const float* __restrict__ x,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
     float* __restrict__ out, int F) {
-  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
+  const int w = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
-  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
-  for (int it = wave; it < n_items; it += n_waves) {
+  const int it_beg = wave_start[w], it_end = wave_start[w + 1];
+  for (int it = it_beg; it < it_end; ++it) {
     int row = wrow[it];
     const bool atomic = row < 0;
     if (atomic) row = ~row;
@@ -473,29 +485,31 @@ int spmm_grid(int n_rows) {
 }
 
 at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
-                    at::Tensor indices, at::Tensor x, int64_t n_rows,
+                    at::Tensor wave_start, at::Tensor indices, at::Tensor x,
+                    int64_t n_rows,
                     c10::optional<at::Tensor> src_scale,
                     c10::optional<at::Tensor> dst_scale,
                     c10::optional<at::Tensor> out_opt) {
   check_f32(x, "x");
   TORCH_CHECK(wrow.scalar_type() == at::kInt &&
                   wbeg.scalar_type() == at::kLong &&
+                  wave_start.scalar_type() == at::kInt &&
                   indices.scalar_type() == at::kInt,
               "worklist int32/int64 + indices int32 expected");
   const int F = x.size(1);
-  const int n_items = wrow.numel();
+  const int n_waves = wave_start.numel() - 1;
   at::Tensor out;
   const bool acc = out_opt.has_value();
   if (acc) { out = *out_opt; check_f32(out, "out"); }
   else { out = at::zeros({n_rows, F}, x.options()); }
-  if (n_items == 0) return out;
+  if (wrow.numel() == 0 || n_waves <= 0) return out;
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int grid = spmm_grid(n_items);
+  const int grid = n_waves / 4;
   if (F % 4 == 0) {
     auto kfn = acc ? spmm_sum_vec4_kernel<true> : spmm_sum_vec4_kernel<false>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
                        wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
-                       wend.data_ptr<int64_t>(), n_items,
+                       wend.data_ptr<int64_t>(), wave_start.data_ptr<int32_t>(),
                        indices.data_ptr<int32_t>(), x.data_ptr<float>(),
                        opt_ptr(src_scale), opt_ptr(dst_scale),
                        out.data_ptr<float>(), F / 4);
@@ -503,7 +517,7 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
     auto kfn = acc ? spmm_sum_scalar_kernel<true> : spmm_sum_scalar_kernel<false>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
                        wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
-                       wend.data_ptr<int64_t>(), n_items,
+                       wend.data_ptr<int64_t>(), wave_start.data_ptr<int32_t>(),
                        indices.data_ptr<int32_t>(), x.data_ptr<float>(),
                        opt_ptr(src_scale), opt_ptr(dst_scale),
                        out.data_ptr<float>(), F);

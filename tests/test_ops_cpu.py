@@ -230,7 +230,7 @@ def test_build_worklist():
     from bnsgcn_amd.ops.csr_torch import build_worklist
     # degrees: 0, 3, 5000, 10 with seg=2048 -> row 2 split into 3 items
     indptr = torch.tensor([0, 0, 3, 5003, 5013], dtype=torch.int64)
-    wrow, wbeg, wend = build_worklist(indptr, seg=2048)
+    wrow, wbeg, wend, wave_start = build_worklist(indptr, seg=2048)
     assert wrow.numel() == 1 + 1 + 3 + 1
     # every edge covered exactly once, rows correct
     cover = torch.zeros(5013, dtype=torch.int32)
@@ -244,6 +244,12 @@ def test_build_worklist():
     for r in wrow.tolist():
         if r < 0:
             assert ~r == 2
-    # longest-first
-    lens = (wend - wbeg).tolist()
-    assert lens == sorted(lens, reverse=True)
+    # wave ranges: monotone, multiple of 4 waves, full coverage
+    assert (wave_start.numel() - 1) % 4 == 0
+    ws = wave_start.tolist()
+    assert ws[0] == 0 and ws[-1] == wrow.numel()
+    assert all(a <= b for a, b in zip(ws, ws[1:]))
+    # edge balance: no wave gets more than one item's worth over the mean
+    lens = (wend - wbeg)
+    per_wave = [int(lens[ws[i]:ws[i + 1]].sum()) for i in range(len(ws) - 1)]
+    assert max(per_wave) <= 5013 / (len(ws) - 1) + 2048
