@@ -68,6 +68,16 @@ def build_worklist(indptr: torch.Tensor, seg: int = SEG, max_waves: int = 65536
     split = nseg[rows] > 1
     wrow = torch.where(split, ~rows, rows).to(torch.int32)
 
+    # drop empty items (deg-0 rows): the launcher zero-initializes out, so
+    # they contribute nothing — and zero-length items degenerate the
+    # edge-balancing searchsorted (every empty item lands on one wave)
+    keep = wend > wbeg
+    wrow, wbeg, wend = wrow[keep], wbeg[keep], wend[keep]
+    total = int(keep.sum())
+    if total == 0:
+        z = torch.zeros(0, dtype=torch.int32, device=device)
+        return (z, wbeg, wend, torch.zeros(1, dtype=torch.int32, device=device))
+
     lens = (wend - wbeg)
     cum = torch.cumsum(torch.cat([torch.zeros(1, dtype=lens.dtype, device=device),
                                   lens]), 0)

@@ -320,7 +320,7 @@ void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
         const int gn = n0 + wc + fj * 16 + j_l;
         if (gm < M && gn < N) {
           float v = acc[fi][fj][r];
-          if (bias) v += bias[gn];
+          if (bias && blockIdx.z == 0) v += bias[gn];  // bias once (slice 0)
           if (ATOMIC) atomicAdd(&C[(int64_t)gm * N + gn], v);
           else C[(int64_t)gm * N + gn] = v;
         }
@@ -681,13 +681,12 @@ at::Tensor gemm_strided(const at::Tensor& A, int64_t sAm, int64_t sAk,
     splitk = std::max(splitk, 1);
   }
   if (splitk > 1) {
-    TORCH_CHECK(!bias.has_value(), "split-K GEMM does not take a bias");
     auto C = at::zeros({M, N}, A.options());
     const int k_slice = ((K + splitk - 1) / splitk + BK - 1) / BK * BK;
     grid.z = (K + k_slice - 1) / k_slice;
     hipLaunchKernelGGL(gemm_f32_kernel<true>, grid, dim3(256), 0, stream,
                        A.data_ptr<float>(), sAm, sAk, B.data_ptr<float>(), sBk,
-                       sBn, nullptr, C.data_ptr<float>(), M, N, K, k_slice);
+                       sBn, opt_ptr(bias), C.data_ptr<float>(), M, N, K, k_slice);
     return C;
   }
   auto C = at::empty({M, N}, A.options());

@@ -231,7 +231,7 @@ def test_build_worklist():
     # degrees: 0, 3, 5000, 10 with seg=2048 -> row 2 split into 3 items
     indptr = torch.tensor([0, 0, 3, 5003, 5013], dtype=torch.int64)
     wrow, wbeg, wend, wave_start = build_worklist(indptr, seg=2048)
-    assert wrow.numel() == 1 + 1 + 3 + 1
+    assert wrow.numel() == 1 + 3 + 1  # deg-0 row dropped
     # every edge covered exactly once, rows correct
     cover = torch.zeros(5013, dtype=torch.int32)
     for r, b, e in zip(wrow.tolist(), wbeg.tolist(), wend.tolist()):
