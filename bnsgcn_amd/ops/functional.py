@@ -205,6 +205,8 @@ class _Linear(Function):
                 gx = e.gemm_nn(grad, weight)
             if ctx.needs_input_grad[1]:
                 gw = e.gemm_tn(grad, x)
+            if ctx.has_bias and ctx.needs_input_grad[2]:
+                return gx, gw, e.syncbn_stats(grad)[0]
         else:
             if ctx.needs_input_grad[0]:
                 gx = grad @ weight

@@ -247,19 +247,20 @@ __global__ void syncbn_stats_kernel(const float* __restrict__ x,
 
 #define BM 64
 #define BN 64
-#define BK 16
+#define BK 32
 
-// ATOMIC variant: split-K — blockIdx.z picks a K slice, partials combine
-// via atomicAdd (used when M/N tiles alone cannot fill 256 CUs, e.g. the
-// dW = g^T @ x reduction GEMMs with K = node count).
+// Double-buffered: while the MFMAs consume LDS buffer `buf`, the next
+// K-tile is loaded into registers and written to buf^1 (one barrier per
+// K-step). Staging thread->element maps follow each operand's unit stride
+// so global reads stay coalesced for every trans layout.
 template <bool ATOMIC>
 __global__ __launch_bounds__(256)
 void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
                      const float* __restrict__ B, int64_t sBk, int64_t sBn,
                      const float* __restrict__ bias, float* __restrict__ C,
                      int M, int N, int K, int k_slice) {
-  __shared__ float As[BM][BK + 1];
-  __shared__ float Bs[BK][BN + 1];
+  __shared__ float As[2][BM][BK + 1];
+  __shared__ float Bs[2][BK][BN + 1];
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
   const int tid = threadIdx.x;
@@ -267,45 +268,67 @@ void gemm_f32_kernel(const float* __restrict__ A, int64_t sAm, int64_t sAk,
   const int wid = tid / WAVE;          // 4 waves: quadrant (wr, wc)
   const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
 
-  using f32x4 = __attribute__((ext_vector_type(4))) float;
-  f32x4 acc[2][2] = {};
-
   const int kb = blockIdx.z * k_slice;
   const int ke = (kb + k_slice < K) ? kb + k_slice : K;
+
+  // per-thread staging coordinates (8 elements of A, 8 of B per K-tile)
+  int ai[8], ak[8], bk[8], bj[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int tA = tid + j * 256;      // over BM*BK = 2048
+    if (sAk == 1) { ai[j] = tA / BK; ak[j] = tA % BK; }
+    else          { ai[j] = tA % BM; ak[j] = tA / BM; }
+    const int tB = tid + j * 256;      // over BK*BN = 2048
+    if (sBn == 1) { bk[j] = tB / BN; bj[j] = tB % BN; }
+    else          { bk[j] = tB % BK; bj[j] = tB / BK; }
+  }
+  float ra[8], rb[8];
+  auto load_regs = [&](int k0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int gm = m0 + ai[j], gka = k0 + ak[j];
+      ra[j] = (gm < M && gka < ke) ? A[gm * sAm + gka * sAk] : 0.f;
+      const int gkb = k0 + bk[j], gn = n0 + bj[j];
+      rb[j] = (gkb < ke && gn < N) ? B[gkb * sBk + gn * sBn] : 0.f;
+    }
+  };
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      As[buf][ai[j]][ak[j]] = ra[j];
+      Bs[buf][bk[j]][bj[j]] = rb[j];
+    }
+  };
+
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  f32x4 acc[2][2] = {};
+  const int i_l = lane & 15, k_l = lane >> 4;
+
+  load_regs(kb);
+  write_lds(0);
+  __syncthreads();
+  int buf = 0;
   for (int k0 = kb; k0 < ke; k0 += BK) {
-    // cooperative staging: 256 threads, 4 elems each, bounds-checked
-    // (zero-fill). The thread->element map follows the operand's unit
-    // stride so global reads stay coalesced for every trans layout.
-    for (int t = tid; t < BM * BK; t += 256) {
-      int i, k;
-      if (sAk == 1) { i = t / BK; k = t % BK; }   // row-major A
-      else          { i = t % BM; k = t / BM; }   // col-major A
-      const int gm = m0 + i, gk = k0 + k;
-      As[i][k] = (gm < M && gk < ke) ? A[gm * sAm + gk * sAk] : 0.f;
-    }
-    for (int t = tid; t < BK * BN; t += 256) {
-      int k, j;
-      if (sBn == 1) { k = t / BN; j = t % BN; }   // row-major B
-      else          { k = t % BK; j = t / BK; }   // col-major B
-      const int gk = k0 + k, gn = n0 + j;
-      Bs[k][j] = (gk < ke && gn < N) ? B[gk * sBk + gn * sBn] : 0.f;
-    }
-    __syncthreads();
+    const bool more = (k0 + BK) < ke;
+    if (more) load_regs(k0 + BK);
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 4) {
-      const int i_l = lane & 15, k_l = lane >> 4;   // fragment lane map
 #pragma unroll
       for (int fi = 0; fi < 2; ++fi) {
-        const float a = As[wr + fi * 16 + i_l][kk + k_l];
+        const float a = As[buf][wr + fi * 16 + i_l][kk + k_l];
 #pragma unroll
         for (int fj = 0; fj < 2; ++fj) {
-          const float b = Bs[kk + k_l][wc + fj * 16 + i_l];
+          const float b = Bs[buf][kk + k_l][wc + fj * 16 + i_l];
           acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b,
                                                              acc[fi][fj], 0, 0, 0);
         }
       }
     }
-    __syncthreads();
+    if (more) {
+      write_lds(buf ^ 1);
+      __syncthreads();
+    }
+    buf ^= 1;
   }
 
   // epilogue: C/D map col=l&15, row=(l>>4)*4+reg
@@ -382,11 +405,11 @@ __global__ void segment_softmax_kernel(const int64_t* __restrict__ indptr,
     for (int64_t e = beg + lane; e < end; e += WAVE)
       m = fmaxf(m, logits[e * H + h]);
     m = wave_reduce_max(m);
-    float s = 0.f;
+    float sum = 0.f;
     for (int64_t e = beg + lane; e < end; e += WAVE)
-      s += __expf(logits[e * H + h] - m);
-    s = wave_reduce_sum(s);
-    const float inv = 1.0f / fmaxf(s, 1e-38f);
+      sum += __expf(logits[e * H + h] - m);
+    sum = wave_reduce_sum(sum);
+    const float inv = 1.0f / fmaxf(sum, 1e-38f);
     for (int64_t e = beg + lane; e < end; e += WAVE)
       alpha[e * H + h] = __expf(logits[e * H + h] - m) * inv;
   }
@@ -404,12 +427,12 @@ __global__ void segment_softmax_bwd_kernel(const int64_t* __restrict__ indptr,
   for (int rh = wave; rh < n_rows * H; rh += n_waves) {
     const int r = rh / H, h = rh % H;
     const int64_t beg = indptr[r], end = indptr[r + 1];
-    float s = 0.f;
+    float sum = 0.f;
     for (int64_t e = beg + lane; e < end; e += WAVE)
-      s += alpha[e * H + h] * grad[e * H + h];
-    s = wave_reduce_sum(s);
+      sum += alpha[e * H + h] * grad[e * H + h];
+    sum = wave_reduce_sum(sum);
     for (int64_t e = beg + lane; e < end; e += WAVE)
-      out[e * H + h] = alpha[e * H + h] * (grad[e * H + h] - s);
+      out[e * H + h] = alpha[e * H + h] * (grad[e * H + h] - sum);
   }
 }
 
