@@ -599,7 +599,8 @@ at::Tensor syncbn_stats(at::Tensor x) {
   const int F = x.size(1);
   auto out = at::zeros({2, F}, x.options());
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int row_chunks = (int)std::min<int64_t>((n + 4095) / 4096, 64);
+  // enough row-chunks to fill 256 CUs even for one 256-wide column block
+  const int row_chunks = (int)std::min<int64_t>((n + 255) / 256, 2048);
   dim3 grid((F + 255) / 256, row_chunks);
   hipLaunchKernelGGL(syncbn_stats_kernel, grid, dim3(256), 0, stream,
                      x.data_ptr<float>(), out.data_ptr<float>(), n, F,
