@@ -35,8 +35,10 @@ def init_distributed(backend: str | None = None, rank: int | None = None,
     os.environ.setdefault("MASTER_PORT", str(master_port))
     if backend is None or backend == "auto":
         backend = "nccl" if torch.cuda.is_available() else "gloo"
+    # generous timeout: rank 0 may be generating + partitioning a
+    # 100M-edge synthetic graph while the others wait at the first barrier
     dist.init_process_group(backend, rank=rank, world_size=world_size,
-                            timeout=datetime.timedelta(seconds=300))
+                            timeout=datetime.timedelta(seconds=1800))
     return rank, world_size
 
 

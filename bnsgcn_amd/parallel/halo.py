@@ -22,6 +22,8 @@ int(p·n) sizes both sides.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 from torch.autograd import Function
 
@@ -58,7 +60,10 @@ class _PartitionAggregate(Function):
     def forward(ctx, x, plan: HaloPlan, st: EpochState,
                 inner_indptr, inner_indices, inner_t_indptr, inner_t_indices,
                 src_scale, dst_scale, use_halo_src_scale: bool):
-        cuda = x.is_cuda
+        # BNSGCN_NO_OVERLAP=1 disables the side-stream overlap (debug /
+        # perf A/B); the sequential branch works for GPU tensors too.
+        cuda = x.is_cuda and os.environ.get("BNSGCN_NO_OVERLAP") != "1"
+
         hscale = st.halo_out_norm_inv if use_halo_src_scale else None
         if cuda:
             cs = comm_stream()
@@ -91,7 +96,7 @@ class _PartitionAggregate(Function):
         inner_t_indptr, inner_t_indices, src_scale, dst_scale = ctx.saved_tensors
         hscale = ctx.hscale
         g = g.contiguous()
-        cuda = g.is_cuda
+        cuda = g.is_cuda and os.environ.get("BNSGCN_NO_OVERLAP") != "1"
         if cuda:
             cs = comm_stream()
             ev = torch.cuda.current_stream().record_event()

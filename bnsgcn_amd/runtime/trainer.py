@@ -36,13 +36,42 @@ from .config import graph_name_of
 
 # --------------------------------------------------------------- offline
 
+def _load_data_cached(args):
+    """Generate the synthetic dataset, caching the edge arrays on disk so
+    back-to-back partitionings for different N (the driver's 1/2/4/8-GPU
+    scaling sweep) skip the expensive edge sampling."""
+    import numpy as np
+    from ..graph.csr import CSR, Graph
+    cache = os.path.join(args.partition_dir,
+                         f"_edges_{args.dataset}_s{args.seed}"
+                         f"_x{args.data_scale:g}.npz")
+    if os.path.exists(cache):
+        z = np.load(cache)
+        adj = CSR(z["indptr"], z["indices"], int(z["n_nodes"]))
+        g = Graph(adj, z["feat"], z["label"], z["train_mask"], z["val_mask"],
+                  z["test_mask"], int(z["n_class"]), bool(z["multilabel"]),
+                  name=args.dataset)
+    else:
+        g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
+        try:
+            os.makedirs(args.partition_dir, exist_ok=True)
+            np.savez(cache, indptr=g.adj_in.indptr, indices=g.adj_in.indices,
+                     n_nodes=g.n_nodes, feat=g.feat, label=g.label,
+                     train_mask=g.train_mask, val_mask=g.val_mask,
+                     test_mask=g.test_mask, n_class=g.n_class,
+                     multilabel=g.multilabel)
+        except OSError:
+            pass
+    return g
+
+
 def prepare_partitions(args) -> str:
     """Rank-0 offline step (reference main.py:26-31 + graph_partition)."""
     name = graph_name_of(args)
     d = os.path.join(args.partition_dir, name)
     if args.skip_partition and os.path.exists(os.path.join(d, "meta.json")):
         return d
-    g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
+    g = _load_data_cached(args)
     if args.inductive:
         g = g.subgraph(g.train_mask, name=g.name)
     extra = {"inductive": args.inductive, "dataset_seed": args.seed,
