@@ -163,8 +163,9 @@ def main():
         baseline = n_edges / ref_epoch_s if a.dataset == "reddit" and \
             a.data_scale == 1.0 else None
         out = {
-            "metric": "training throughput (full-graph edges/s), Reddit-shaped "
-                      "GraphSAGE 3-layer h=256, sampling-rate=0.1",
+            "metric": f"training throughput (full-graph edges/s), "
+                      f"{a.dataset}-shaped {a.model} {a.n_layers}-layer "
+                      f"h={a.n_hidden}, sampling-rate={a.sampling_rate}",
             "value": value,
             "unit": "edges/s",
             "n_gpus": world,
