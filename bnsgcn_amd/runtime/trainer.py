@@ -271,6 +271,8 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
     loss_history: list[float] = []
     loss_val = float("nan")
     cuda = str(device).startswith("cuda")
+    if cuda:
+        torch.cuda.reset_peak_memory_stats()
 
     for epoch in range(args.n_epochs):
         t0 = time.perf_counter()
@@ -313,6 +315,13 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
             pending = pool.submit(lambda m=model_cpu, s=snap:
                                   (evaluator.evaluate(m), s))
 
+    if cuda:
+        # reference print_memory (helper/utils.py:244-250, train.py:444)
+        print(f"Process {rank:03d} | "
+              f"allocated {torch.cuda.memory_allocated() / 2**20:.0f} MB | "
+              f"peak {torch.cuda.max_memory_allocated() / 2**20:.0f} MB | "
+              f"reserved {torch.cuda.memory_reserved() / 2**20:.0f} MB",
+              flush=True)
     summary = {"rank": rank, "epoch_time": float(np.mean(train_dur)) if train_dur
                else float("nan"),
                "comm_time": float(np.mean(comm_dur)) if comm_dur else 0.0,
