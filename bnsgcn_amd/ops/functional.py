@@ -38,13 +38,18 @@ def spmm_sum_raw(indptr, indices, x, src_scale=None, dst_scale=None, out=None):
 
 def spmm_edge_raw(indptr, indices, eweight, x, out=None):
     if use_hip(x):
-        return get_ext().spmm_edge_sum(indptr, indices, eweight, x, out)
+        wl = _worklist_of(indptr)
+        return get_ext().spmm_edge_sum(*wl, indptr, indices,
+                                       eweight.contiguous(), x.contiguous(),
+                                       out)
     return ref.spmm_edge_sum(indptr, indices, eweight, x, out)
 
 
 def sddmm_dot_raw(indptr, indices, a_dst, b_src):
     if use_hip(a_dst):
-        return get_ext().sddmm_dot(indptr, indices, a_dst, b_src)
+        wl = _worklist_of(indptr)
+        return get_ext().sddmm_dot(*wl, indptr, indices,
+                                   a_dst.contiguous(), b_src.contiguous())
     return ref.sddmm_dot(indptr, indices, a_dst, b_src)
 
 
@@ -145,10 +150,18 @@ class _SDDMMAdd(Function):
         indptr, indices, indptr_t, indices_t, eperm_t = ctx.saved_tensors
         grad = grad.contiguous()
         g_el = g_er = None
-        ones = None
+        if use_hip(grad) and grad.shape[1] <= 8:
+            e = get_ext()
+            if ctx.needs_input_grad[0]:
+                # d el[c] = Σ_{e: col_e=c} grad[e]: segment-sum over the
+                # transposed CSR with the edge permutation
+                g_el = e.segment_sum_edges(*_worklist_of(indptr_t), eperm_t,
+                                           grad, ctx.n_src)
+            if ctx.needs_input_grad[1]:
+                g_er = e.segment_sum_edges(*_worklist_of(indptr), None,
+                                           grad, ctx.n_dst)
+            return g_el, g_er, None, None, None, None, None
         if ctx.needs_input_grad[0]:
-            # d el[c] = Σ_{e: col_e=c} grad[e]: edge-weighted spmm on the
-            # transpose with x = ones → equivalently segment-sum by col.
             g_el = torch.zeros(ctx.n_src, grad.shape[1], dtype=grad.dtype,
                                device=grad.device)
             g_el.index_add_(0, indices.long(), grad)
