@@ -463,27 +463,39 @@ __global__ void spmm_edge_kernel(const int64_t* __restrict__ indptr,
   }
 }
 
-// gw[e,h] = <g[row_e,h,:], x[col_e,h,:]> — wave per edge, lanes over D.
-__global__ void sddmm_dot_kernel(const int64_t* __restrict__ indptr,
-                                 const int32_t* __restrict__ indices,
-                                 const float* __restrict__ g,
-                                 const float* __restrict__ x,
-                                 float* __restrict__ out,
-                                 int n_rows, int64_t n_edges, int H, int D) {
-  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+// gw[e,h] = <g[row_e,h,:], x[col_e,h,:]>, general-D form: work-list
+// scheduled, batched coalesced index loads, full-wave shfl reduction per
+// (edge, head). Used when the register-resident fast form's shape
+// conditions fail (e.g. D % 8 != 0 on the GAT output layer).
+__global__ __launch_bounds__(256) void sddmm_dot_gen_kernel(
+    const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
+    const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
+    const int32_t* __restrict__ indices, const float* __restrict__ g,
+    const float* __restrict__ x, float* __restrict__ out, int H, int D,
+    int HD) {
+  const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
+  const int wv = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
-  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
-  for (int r = wave; r < n_rows; r += n_waves) {
-    const int64_t beg = indptr[r], end = indptr[r + 1];
-    for (int64_t e = beg; e < end; ++e) {
-      const int c = indices[e];
-      for (int h = 0; h < H; ++h) {
-        float acc = 0.f;
-        for (int d = lane; d < D; d += WAVE)
-          acc += g[((int64_t)r * H + h) * D + d] *
-                 x[((int64_t)c * H + h) * D + d];
-        acc = wave_reduce_sum(acc);
-        if (lane == 0) out[e * H + h] = acc;
+  const int it_beg = wave_start[wv], it_end = wave_start[wv + 1];
+  for (int it = it_beg; it < it_end; ++it) {
+    int row = wrow[it];
+    if (row < 0) row = ~row;
+    const int64_t beg = wbeg[it], end = wend[it];
+    for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
+      const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
+      int cid = 0;
+      if (lane < nv) cid = indices[e0 + lane];
+#pragma unroll 2
+      for (int k = 0; k < nv; ++k) {
+        const int c = __shfl(cid, k, WAVE);
+        for (int h = 0; h < H; ++h) {
+          float p = 0.f;
+          for (int d = lane; d < D; d += WAVE)
+            p += g[(int64_t)row * HD + h * D + d] *
+                 x[(int64_t)c * HD + h * D + d];
+          p = wave_reduce_sum(p);
+          if (lane == 0) out[(e0 + k) * H + h] = p;
+        }
       }
     }
   }
@@ -856,11 +868,11 @@ at::Tensor sddmm_dot(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                        out.data_ptr<float>(), H, D, HD);
     return out;
   }
-  hipLaunchKernelGGL(sddmm_dot_kernel, dim3(spmm_grid(n_rows)), dim3(256), 0,
-                     stream, indptr.data_ptr<int64_t>(),
+  hipLaunchKernelGGL(sddmm_dot_gen_kernel, dim3(n_waves / 4), dim3(256), 0,
+                     stream, wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
+                     wend.data_ptr<int64_t>(), wstart.data_ptr<int32_t>(),
                      indices.data_ptr<int32_t>(), g.data_ptr<float>(),
-                     x.data_ptr<float>(), out.data_ptr<float>(), n_rows,
-                     indices.numel(), H, D);
+                     x.data_ptr<float>(), out.data_ptr<float>(), H, D, H * D);
   return out;
 }
 
