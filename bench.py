@@ -44,6 +44,7 @@ def parse():
     p.add_argument("--n-layers", type=int, default=3)
     p.add_argument("--n-hidden", type=int, default=256)
     p.add_argument("--heads", type=int, default=4)
+    p.add_argument("--n-linear", type=int, default=0)
     p.add_argument("--sampling-rate", type=float, default=0.1)
     p.add_argument("--partition-method", type=str, default="metis")
     p.add_argument("--data-scale", type=float, default=1.0)
@@ -70,6 +71,7 @@ def main():
     args.n_layers = a.n_layers
     args.n_hidden = a.n_hidden
     args.heads = a.heads
+    args.n_linear = a.n_linear
     args.sampling_rate = a.sampling_rate
     args.n_partitions = world
     args.partition_method = a.partition_method
