@@ -25,7 +25,14 @@ def transpose_csr(indptr: torch.Tensor, indices: torch.Tensor, n_cols: int
                                   row_lengths(indptr))
     eperm = torch.argsort(indices.long(), stable=True)
     indices_t = row[eperm]
-    counts = torch.bincount(indices.long(), minlength=n_cols)
+    if indices.is_cuda and indices.dtype == torch.int32:
+        from ._ext import get_ext, has_ext
+        if has_ext():
+            counts = get_ext().bincount_i32(indices, n_cols)
+        else:
+            counts = torch.bincount(indices.long(), minlength=n_cols)
+    else:
+        counts = torch.bincount(indices.long(), minlength=n_cols)
     indptr_t = torch.zeros(n_cols + 1, dtype=indptr.dtype, device=device)
     torch.cumsum(counts, 0, out=indptr_t[1:])
     return indptr_t, indices_t, eperm

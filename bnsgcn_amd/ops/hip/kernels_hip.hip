@@ -651,6 +651,16 @@ __global__ __launch_bounds__(256) void segment_sum_edges_kernel(
   }
 }
 
+// int32 bincount (torch's histogram kernel measured 6.8 ms on a 14M-edge
+// per-epoch transpose; this is a plain atomic histogram, ~0.1 ms)
+__global__ void bincount_i32_kernel(const int32_t* __restrict__ v, int64_t n,
+                                    int64_t* __restrict__ out) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = t; i < n; i += stride)
+    atomicAdd(reinterpret_cast<unsigned long long*>(&out[v[i]]), 1ull);
+}
+
 // ------------------------------ launchers ------------------------------
 
 inline void check_f32(const at::Tensor& t, const char* name) {
@@ -767,6 +777,17 @@ at::Tensor syncbn_stats(at::Tensor x) {
   hipLaunchKernelGGL(syncbn_stats_kernel, grid, dim3(256), 0, stream,
                      x.data_ptr<float>(), out.data_ptr<float>(), n, F,
                      row_chunks);
+  return out;
+}
+
+at::Tensor bincount_i32(at::Tensor v, int64_t n_bins) {
+  TORCH_CHECK(v.is_cuda() && v.scalar_type() == at::kInt, "int32 cuda input");
+  auto out = at::zeros({n_bins}, v.options().dtype(at::kLong));
+  if (v.numel() == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int grid = std::min<int64_t>((v.numel() + 255) / 256, 4096);
+  hipLaunchKernelGGL(bincount_i32_kernel, dim3(grid), dim3(256), 0, stream,
+                     v.data_ptr<int32_t>(), v.numel(), out.data_ptr<int64_t>());
   return out;
 }
 
@@ -966,6 +987,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "edge softmax backward");
   m.def("spmm_edge_sum", &spmm_edge_sum, "multi-head edge-weighted SpMM");
   m.def("sddmm_dot", &sddmm_dot, "per-edge per-head dot (spmm_edge grad)");
+  m.def("bincount_i32", &bincount_i32, "atomic int32 histogram");
   m.def("segment_sum_edges", &segment_sum_edges,
         "segment sum of (permuted) edge values by CSR row");
 }
