@@ -36,13 +36,16 @@ def spmm_sum_raw(indptr, indices, x, src_scale=None, dst_scale=None, out=None):
     return ref.spmm_sum(indptr, indices, x, src_scale, dst_scale, out)
 
 
-def spmm_edge_raw(indptr, indices, eweight, x, out=None):
+def spmm_edge_raw(indptr, indices, eweight, x, out=None, wperm=None):
+    """wperm: optional edge permutation applied to eweight INSIDE the HIP
+    kernel (fuses the w[eperm] gather of the transposed backward pass)."""
     if use_hip(x):
         wl = _worklist_of(indptr)
         return get_ext().spmm_edge_sum(*wl, indptr, indices,
-                                       eweight.contiguous(), x.contiguous(),
-                                       out)
-    return ref.spmm_edge_sum(indptr, indices, eweight, x, out)
+                                       eweight.contiguous(), wperm,
+                                       x.contiguous(), out)
+    ew = eweight if wperm is None else eweight[wperm]
+    return ref.spmm_edge_sum(indptr, indices, ew, x, out)
 
 
 def sddmm_dot_raw(indptr, indices, a_dst, b_src):
@@ -126,7 +129,7 @@ class _SpMMEdge(Function):
         grad = grad.contiguous()
         gx = gw = None
         if ctx.needs_input_grad[0]:
-            gx = spmm_edge_raw(indptr_t, indices_t, w[eperm_t], grad)
+            gx = spmm_edge_raw(indptr_t, indices_t, w, grad, wperm=eperm_t)
         if ctx.needs_input_grad[1]:
             gw = sddmm_dot_raw(indptr, indices, grad, x)
         return gx, gw, None, None, None, None, None

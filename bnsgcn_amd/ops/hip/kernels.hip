@@ -477,6 +477,8 @@ __global__ __launch_bounds__(256) void sddmm_dot_gen_kernel(
   const int wv = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
   const int it_beg = wave_start[wv], it_end = wave_start[wv + 1];
+  const int sg = lane >> 4;            // 4 subgroups of 16 lanes:
+  const int gl = lane & 15;            // 4 (edge) pairs in flight per wave
   for (int it = it_beg; it < it_end; ++it) {
     int row = wrow[it];
     if (row < 0) row = ~row;
@@ -485,16 +487,19 @@ __global__ __launch_bounds__(256) void sddmm_dot_gen_kernel(
       const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
       int cid = 0;
       if (lane < nv) cid = indices[e0 + lane];
-#pragma unroll 2
-      for (int k = 0; k < nv; ++k) {
-        const int c = __shfl(cid, k, WAVE);
+      for (int k0 = 0; k0 < nv; k0 += 4) {
+        const int k = k0 + sg;
+        const bool act = k < nv;
+        const int c = __shfl(cid, act ? k : 0, WAVE);
         for (int h = 0; h < H; ++h) {
           float p = 0.f;
-          for (int d = lane; d < D; d += WAVE)
-            p += g[(int64_t)row * HD + h * D + d] *
-                 x[(int64_t)c * HD + h * D + d];
-          p = wave_reduce_sum(p);
-          if (lane == 0) out[(e0 + k) * H + h] = p;
+          if (act)
+            for (int d = gl; d < D; d += 16)
+              p += g[(int64_t)row * HD + h * D + d] *
+                   x[(int64_t)c * HD + h * D + d];
+#pragma unroll
+          for (int off = 8; off > 0; off >>= 1) p += __shfl_xor(p, off, WAVE);
+          if (act && gl == 0) out[(e0 + k) * H + h] = p;
         }
       }
     }
@@ -512,8 +517,8 @@ __global__ __launch_bounds__(256) void spmm_edge_vec4_kernel(
     const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
     const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
     const int32_t* __restrict__ indices, const float* __restrict__ w,
-    const float* __restrict__ x, float* __restrict__ out, int H, int D4,
-    int HD4) {
+    const int64_t* __restrict__ wperm, const float* __restrict__ x,
+    float* __restrict__ out, int H, int D4, int HD4) {
   const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
   const int wv = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
@@ -538,7 +543,8 @@ __global__ __launch_bounds__(256) void spmm_edge_vec4_kernel(
         for (int k = 0; k < nv; ++k) {
           const int c = __shfl(cid, k, WAVE);
           if (hasA) {
-            const float ww = w[(e0 + k) * H + hA];
+            const int64_t we = wperm ? wperm[e0 + k] : (e0 + k);
+            const float ww = w[we * H + hA];
             f4_axpy(acc, ww, x4[(int64_t)c * HD4 + fA]);
           }
         }
@@ -836,7 +842,8 @@ at::Tensor segment_softmax_backward(at::Tensor indptr, at::Tensor alpha,
 
 at::Tensor spmm_edge_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                          at::Tensor wstart, at::Tensor indptr,
-                         at::Tensor indices, at::Tensor w, at::Tensor x,
+                         at::Tensor indices, at::Tensor w,
+                         c10::optional<at::Tensor> wperm, at::Tensor x,
                          c10::optional<at::Tensor> out_opt) {
   check_f32(w, "w"); check_f32(x, "x");
   const int n_rows = indptr.numel() - 1;
@@ -854,10 +861,13 @@ at::Tensor spmm_edge_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                        wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
                        wend.data_ptr<int64_t>(), wstart.data_ptr<int32_t>(),
                        indices.data_ptr<int32_t>(), w.data_ptr<float>(),
+                       wperm.has_value() ? wperm->data_ptr<int64_t>() : nullptr,
                        x.data_ptr<float>(), out.data_ptr<float>(), H, D / 4,
                        H * D / 4);
     return out;
   }
+  at::Tensor wp = wperm.has_value() ? w.index_select(0, *wperm) : w;
+  w = wp;
   auto kfn = acc ? spmm_edge_kernel<true> : spmm_edge_kernel<false>;
   hipLaunchKernelGGL(kfn, dim3(spmm_grid(n_rows * H)), dim3(256), 0, stream,
                      indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
