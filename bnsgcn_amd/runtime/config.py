@@ -55,6 +55,10 @@ def create_parser() -> argparse.ArgumentParser:
                    help="reproduce the reference's (acknowledged-wrong, "
                         "train.py:117) 1/ratio scaling of GAT attention inputs; "
                         "default keeps ratio=1 for GAT")
+    p.add_argument("--eval-device", type=str, default="cpu",
+                   help="device for rank-0 full-graph evaluation (the "
+                        "reference evaluates on CPU; 'cuda' runs it on the "
+                        "rank-0 GPU with the HIP kernels)")
     p.add_argument("--bucket-mb", type=int, default=16,
                    help="gradient all-reduce bucket size (MiB)")
     return p

@@ -163,6 +163,7 @@ class Evaluator:
 
     def __init__(self, args):
         self.args = args
+        self.device = torch.device(getattr(args, "eval_device", "cpu"))
         self.multilabel = None
         g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
         self.multilabel = g.multilabel
@@ -176,36 +177,38 @@ class Evaluator:
     def _ctx_and_tensors(self, g):
         ctx = GraphContext.for_full_graph(
             torch.from_numpy(g.adj_in.indptr), torch.from_numpy(g.adj_in.indices),
-            torch.from_numpy(g.in_deg), torch.from_numpy(g.out_deg), "cpu")
-        feat = torch.from_numpy(g.feat)
-        label = torch.from_numpy(g.label)
+            torch.from_numpy(g.in_deg), torch.from_numpy(g.out_deg), self.device)
+        feat = torch.from_numpy(g.feat).to(self.device)
+        label = torch.from_numpy(g.label).to(self.device)
         return ctx, feat, label, g
 
     @torch.no_grad()
     def evaluate(self, model_cpu) -> dict:
+        model_cpu = model_cpu.to(self.device)
         model_cpu.eval()
         out = {}
         if self.args.inductive:
             ctx, feat, label, g = self._ctx_and_tensors(self._graphs["val"])
             logits = model_cpu(ctx, feat)
-            vm = torch.from_numpy(g.val_mask)
+            vm = torch.from_numpy(g.val_mask).to(self.device)
             out["val"] = _accuracy(logits[vm], label[vm], self.multilabel)
         else:
             ctx, feat, label, g = self._ctx_and_tensors(self._graphs["full"])
             logits = model_cpu(ctx, feat)
-            vm = torch.from_numpy(g.val_mask)
-            tm = torch.from_numpy(g.test_mask)
+            vm = torch.from_numpy(g.val_mask).to(self.device)
+            tm = torch.from_numpy(g.test_mask).to(self.device)
             out["val"] = _accuracy(logits[vm], label[vm], self.multilabel)
             out["test"] = _accuracy(logits[tm], label[tm], self.multilabel)
         return out
 
     @torch.no_grad()
     def evaluate_test(self, model_cpu) -> float:
+        model_cpu = model_cpu.to(self.device)
         model_cpu.eval()
         key = "test" if self.args.inductive else "full"
         ctx, feat, label, g = self._ctx_and_tensors(self._graphs[key])
         logits = model_cpu(ctx, feat)
-        tm = torch.from_numpy(g.test_mask)
+        tm = torch.from_numpy(g.test_mask).to(self.device)
         return _accuracy(logits[tm], label[tm], self.multilabel)
 
 

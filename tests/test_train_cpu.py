@@ -131,11 +131,14 @@ def test_eval_and_checkpoint(tmp_path):
     os.chdir(tmp_path)
     try:
         multi = _run_config(tmp_path, 2, model="graphsage", sampling_rate=0.5,
-                            use_pp=True, n_epochs=10, log_every=5, eval=True)
+                            use_pp=True, n_epochs=300, log_every=50, eval=True,
+                            lr=0.05, n_hidden=32)
         assert "test_acc" in multi[0]
-        assert 0.0 <= multi[0]["test_acc"] <= 1.0
+        # labels are learnable (synthetic feature-linked): well above the
+        # 1/7 chance level even with this short run
+        assert multi[0]["test_acc"] > 0.30
         name = "tiny-2-metis-vol-trans"
-        assert os.path.exists(f"checkpoint/{name}_p0.50_4.pth.tar")
+        assert os.path.exists(f"checkpoint/{name}_p0.50_49.pth.tar")
         assert os.path.exists(f"checkpoint/{name}_final.pth.tar")
         assert os.path.exists("results/tiny_n2_p0.50.txt")
     finally:
