@@ -18,6 +18,8 @@ from dataclasses import dataclass
 
 import numpy as np
 
+import zlib
+
 from .csr import CSR, Graph, add_self_loops
 
 
@@ -95,7 +97,12 @@ def load_data(name: str, seed: int = 0, scale: float = 1.0) -> Graph:
     spec = DATASETS[name]
     n_nodes = max(16, int(spec.n_nodes * scale))
     n_edges = max(64, int(spec.n_edges * scale))
-    rng = np.random.default_rng(np.random.SeedSequence([hash(name) & 0x7FFFFFFF, seed]))
+    # NOTE: deterministic name hash — python's hash() is randomized per
+    # process (PYTHONHASHSEED), which would make every process regenerate a
+    # DIFFERENT graph (a bug this caused: rank-0's evaluator scored a model
+    # against a different random dataset than it trained on)
+    rng = np.random.default_rng(
+        np.random.SeedSequence([zlib.crc32(name.encode()) & 0x7FFFFFFF, seed]))
 
     src, dst = _draw_edges(spec, n_nodes, n_edges, rng)
     src, dst = add_self_loops(src, dst, n_nodes)
