@@ -45,3 +45,51 @@ def test_boundary_discovery_matches_store():
              "halo_out_deg": p.halo_out_deg} for p in parts]
     res = run_dist(2, _check_rank, (data,))
     assert all(res)
+
+
+def _syncbn_rank(rank, world, xs):
+    from bnsgcn_amd.parallel import init_distributed
+    from bnsgcn_amd.models.sync_bn import SyncBatchNorm
+    import torch
+    init_distributed("gloo", rank, world)
+    torch.manual_seed(0)
+    bn = SyncBatchNorm(xs[0].shape[1], whole_size=sum(x.shape[0] for x in xs))
+    bn.train()
+    x = xs[rank].clone().requires_grad_(True)
+    y = bn(x)
+    g = torch.ones_like(y) * (rank + 1)
+    y.backward(g.clone())
+    # return numpy copies: torch tensors through an mp queue travel as
+    # shared-memory handles that vanish when the child exits
+    return {k: v.detach().numpy().copy() for k, v in
+            {"y": y, "dx": x.grad, "dw": bn.weight.grad, "db": bn.bias.grad,
+             "rm": bn.running_mean, "rv": bn.running_var}.items()}
+
+
+def test_syncbn_matches_single_process_batchnorm():
+    """2-rank SyncBN == torch BatchNorm1d over the concatenated rows
+    (reference module/sync_bn.py semantics with whole_size = total rows)."""
+    import torch
+    torch.manual_seed(3)
+    xs = [torch.randn(40, 6), torch.randn(24, 6)]
+    res = run_dist(2, _syncbn_rank, (xs,))
+    xfull = torch.cat(xs).requires_grad_(True)
+    bn = torch.nn.BatchNorm1d(6, eps=1e-5, momentum=0.1)
+    bn.train()
+    yfull = bn(xfull)
+    gfull = torch.cat([torch.ones_like(xs[0]) * 1, torch.ones_like(xs[1]) * 2])
+    yfull.backward(gfull)
+    res = [{k: torch.from_numpy(v) for k, v in r.items()} for r in res]
+    ycat = torch.cat([res[0]["y"], res[1]["y"]])
+    torch.testing.assert_close(ycat, yfull.detach(), rtol=1e-4, atol=1e-5)
+    dxcat = torch.cat([res[0]["dx"], res[1]["dx"]])
+    torch.testing.assert_close(dxcat, xfull.grad, rtol=1e-4, atol=1e-5)
+    # weight/bias grads are all-reduced: same on both ranks, equal to full
+    torch.testing.assert_close(res[0]["dw"], bn.weight.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(res[0]["db"], bn.bias.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(res[0]["rm"], bn.running_mean, rtol=1e-4, atol=1e-5)
+    # running_var: torch tracks the UNBIASED variance; the reference's
+    # SyncBN (module/sync_bn.py:21) and ours track the biased one
+    n = xfull.shape[0]
+    ours_unbiased = (res[0]["rv"] - 0.9) * n / (n - 1) + 0.9  # undo momentum mix
+    torch.testing.assert_close(ours_unbiased, bn.running_var, rtol=1e-3, atol=1e-4)
