@@ -55,6 +55,10 @@ def create_parser() -> argparse.ArgumentParser:
                    help="reproduce the reference's (acknowledged-wrong, "
                         "train.py:117) 1/ratio scaling of GAT attention inputs; "
                         "default keeps ratio=1 for GAT")
+    p.add_argument("--resume", type=str, default="",
+                   help="checkpoint (.pth.tar state_dict) to load before "
+                        "training — the reference saves checkpoints but has "
+                        "no resume path (SURVEY.md §5.4); this adds one")
     p.add_argument("--eval-device", type=str, default="cpu",
                    help="device for rank-0 full-graph evaluation (the "
                         "reference evaluates on CPU; 'cuda' runs it on the "

