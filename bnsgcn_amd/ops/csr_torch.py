@@ -34,10 +34,13 @@ def transpose_csr(indptr: torch.Tensor, indices: torch.Tensor, n_cols: int
     return indptr_t, indices_t, eperm
 
 
-SEG = 2048  # max edges per SpMM work item (heavy-row split granularity)
+import os
+
+SEG = int(os.environ.get("BNSGCN_SEG", 2048))        # heavy-row split granularity
+MAX_WAVES = int(os.environ.get("BNSGCN_MAX_WAVES", 65536))
 
 
-def build_worklist(indptr: torch.Tensor, seg: int = SEG, max_waves: int = 65536
+def build_worklist(indptr: torch.Tensor, seg: int = None, max_waves: int = None
                    ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
     """Edge-balanced work schedule for the gfx950 SpMM kernel.
 
@@ -56,6 +59,8 @@ def build_worklist(indptr: torch.Tensor, seg: int = SEG, max_waves: int = 65536
     int64, wave_start int32 [n_waves+1]); n_waves is a multiple of 4
     (one 256-thread block = 4 waves).
     """
+    seg = seg or SEG
+    max_waves = max_waves or MAX_WAVES
     device = indptr.device
     deg = (indptr[1:] - indptr[:-1])
     n = deg.numel()
