@@ -253,3 +253,28 @@ def test_build_worklist():
     lens = (wend - wbeg)
     per_wave = [int(lens[ws[i]:ws[i + 1]].sum()) for i in range(len(ws) - 1)]
     assert max(per_wave) <= 5013 / (len(ws) - 1) + 2048
+
+
+def test_gradcheck_custom_functions_f64():
+    """torch.autograd.gradcheck (float64) on every custom Function."""
+    from torch.autograd import gradcheck
+    torch.manual_seed(12)
+    indptr, indices = rand_csr(8, 10, 30, seed=12)
+    indptr_t, indices_t, eperm_t = transpose_csr(indptr, indices, 10)
+    ss = (torch.rand(10) + 0.5).double()
+    ds = (torch.rand(8) + 0.5).double()
+    x = torch.randn(10, 3, dtype=torch.float64, requires_grad=True)
+    assert gradcheck(lambda t: F.spmm_sum(t, indptr, indices, indptr_t,
+                                          indices_t, ss, ds), (x,))
+    H, D = 2, 3
+    xe = torch.randn(10, H, D, dtype=torch.float64, requires_grad=True)
+    w = torch.randn(30, H, dtype=torch.float64, requires_grad=True)
+    assert gradcheck(lambda a, b: F.spmm_edge_sum(a, b, indptr, indices,
+                                                  indptr_t, indices_t, eperm_t),
+                     (xe, w))
+    el = torch.randn(10, H, dtype=torch.float64, requires_grad=True)
+    er = torch.randn(8, H, dtype=torch.float64, requires_grad=True)
+    assert gradcheck(lambda a, b: F.sddmm_add(a, b, indptr, indices, indptr_t,
+                                              indices_t, eperm_t), (el, er))
+    lg = torch.randn(30, H, dtype=torch.float64, requires_grad=True)
+    assert gradcheck(lambda t: F.segment_softmax(t, indptr), (lg,))
