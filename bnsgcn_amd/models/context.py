@@ -36,10 +36,13 @@ class GraphContext:
         self.plan = plan
         self.indptr = indptr.to(dev)
         self.indices = indices.to(dev)
-        tip, tix, _ = transpose_csr(self.indptr, self.indices,
-                                    n_cols=self.indptr.numel() - 1
-                                    if plan is None else plan.n_inner)
+        tip, tix, t_eperm = transpose_csr(self.indptr, self.indices,
+                                          n_cols=self.indptr.numel() - 1
+                                          if plan is None else plan.n_inner)
         self.t_indptr, self.t_indices = tip, tix
+        # t_eperm[j] = original edge id of the j-th transposed edge (for
+        # per-edge payloads through the transpose — GAT attention weights)
+        self.t_eperm = t_eperm
         in_deg = in_deg.to(dev).float()
         out_deg = out_deg.to(dev).float()
         self.in_norm_inv = _inv_sqrt(in_deg)       # GCN dst scale
@@ -129,6 +132,28 @@ class GraphContext:
         tip, tix, eperm = transpose_csr(ip, ix, n_inner + n_halo)
         self._gat_full_cache = (ip, ix, tip, tix, eperm, n_halo)
         return self._gat_full_cache
+
+    def gat_split_halo(self):
+        """Per-epoch sampled-halo edge set for the SPLIT GAT block:
+        (fwd ip/ix rows=inner dst cols=recv idx, bwd ip/ix, eperm_t).
+        Together with the STATIC inner CSRs this replaces the per-epoch
+        merged block (no 14M-edge merge/transpose per epoch)."""
+        st = self.plan.state
+        return (st.halo_fwd_indptr, st.halo_fwd_indices,
+                st.halo_bwd_indptr, st.halo_bwd_indices, st.halo_eperm_t)
+
+    def gat_split_full(self):
+        """Static FULL-halo edge set (GAT layer 0 under use_pp)."""
+        if getattr(self, "_gat_split_full_cache", None) is None:
+            fip, fix, eperm_f = transpose_csr(self.plan.halo_indptr,
+                                              self.plan.halo_indices,
+                                              self.plan.n_inner)
+            eperm_t = torch.empty_like(eperm_f)
+            eperm_t[eperm_f] = torch.arange(eperm_f.numel(),
+                                            device=eperm_f.device)
+            self._gat_split_full_cache = (fip, fix, self.plan.halo_indptr,
+                                          self.plan.halo_indices, eperm_t)
+        return self._gat_split_full_cache
 
     # -------------------------------------------------- full-halo exchange
     def full_state(self) -> EpochState:

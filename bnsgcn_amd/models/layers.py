@@ -102,31 +102,53 @@ class GATLayer(nn.Module):
         """x: [n_local, F]. In partition mode, halo sources are fetched via
         halo_exchange — except when `halo_feat` is given (GAT layer 0 under
         use_pp: the FULL unsampled halo features captured at precompute,
-        reference model.py:118-120 / train.py:208-209)."""
+        reference model.py:118-120 / train.py:208-209).
+
+        Partition mode uses the SPLIT block: attention logits are computed
+        over the static inner edge set and the per-epoch sampled-halo edge
+        set separately, normalized jointly with the union softmax
+        (ops segment_softmax2) — no per-epoch merge/transpose of the big
+        combined CSR. Evaluation (full graph, no plan) keeps the single
+        combined path."""
         from ..parallel.halo import halo_exchange
 
         h = self.feat_drop(x)
         H, D = self.heads, self.out_feats
-        if ctx.plan is not None:
-            if halo_feat is not None:
-                src_extra = self.feat_drop(halo_feat)
-                ip, ix, tip, tix, eperm, _ = ctx.gat_block_full()
-            else:
-                src_extra = halo_exchange(h, ctx.plan)
-                ip, ix, tip, tix, eperm, _ = ctx.gat_block()
-            src = torch.cat((h, src_extra), dim=0)
-        else:
-            src = h
+        if ctx.plan is None:                       # full-graph eval path
             ip, ix, tip, tix, eperm, _ = ctx.gat_block()
-        z_src = F.linear(src, self.fc).view(-1, H, D)
-        z_dst = z_src[:x.shape[0]]
-        el = (z_src * self.attn_l).sum(-1)          # [n_src, H]
-        er = (z_dst * self.attn_r).sum(-1)          # [n_dst, H]
-        logits = F.sddmm_add(el, er, ip, ix, tip, tix, eperm)
-        logits = torch.nn.functional.leaky_relu(logits, self.negative_slope)
-        alpha = F.segment_softmax(logits, ip)
-        alpha = self.attn_drop(alpha)
-        out = F.spmm_edge_sum(z_src, alpha, ip, ix, tip, tix, eperm)
+            z = F.linear(h, self.fc).view(-1, H, D)
+            el = (z * self.attn_l).sum(-1)
+            er = (z * self.attn_r).sum(-1)
+            logits = F.sddmm_add(el, er, ip, ix, tip, tix, eperm)
+            logits = torch.nn.functional.leaky_relu(logits, self.negative_slope)
+            alpha = self.attn_drop(F.segment_softmax(logits, ip))
+            out = F.spmm_edge_sum(z, alpha, ip, ix, tip, tix, eperm)
+            return out + self.bias.view(1, H, D) if self.bias is not None else out
+
+        if halo_feat is not None:                  # layer 0 under use_pp
+            src_halo = self.feat_drop(halo_feat)
+            hip, hix, hbip, hbix, heperm = ctx.gat_split_full()
+        else:
+            src_halo = halo_exchange(h, ctx.plan)
+            hip, hix, hbip, hbix, heperm = ctx.gat_split_halo()
+
+        z_in = F.linear(h, self.fc).view(-1, H, D)
+        z_h = F.linear(src_halo, self.fc).view(-1, H, D)
+        el_in = (z_in * self.attn_l).sum(-1)       # inner sources
+        el_h = (z_h * self.attn_l).sum(-1)         # halo sources
+        er = (z_in * self.attn_r).sum(-1)          # inner destinations
+        li = F.sddmm_add(el_in, er, ctx.indptr, ctx.indices,
+                         ctx.t_indptr, ctx.t_indices, ctx.t_eperm)
+        lh = F.sddmm_add(el_h, er, hip, hix, hbip, hbix, heperm)
+        slope = self.negative_slope
+        li = torch.nn.functional.leaky_relu(li, slope)
+        lh = torch.nn.functional.leaky_relu(lh, slope)
+        ai, ah = F.segment_softmax2(li, lh, ctx.indptr, hip)
+        ai = self.attn_drop(ai)
+        ah = self.attn_drop(ah)
+        out = F.spmm_edge_sum(z_in, ai, ctx.indptr, ctx.indices,
+                              ctx.t_indptr, ctx.t_indices, ctx.t_eperm)
+        out = out + F.spmm_edge_sum(z_h, ah, hip, hix, hbip, hbix, heperm)
         if self.bias is not None:
             out = out + self.bias.view(1, H, D)
         return out

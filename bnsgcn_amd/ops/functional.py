@@ -235,3 +235,37 @@ class _Linear(Function):
 
 def linear(x, weight, bias=None):
     return _Linear.apply(x, weight, bias)
+
+
+def segment_softmax2_raw(indptr1, logits1, indptr2, logits2):
+    if use_hip(logits1):
+        return get_ext().segment_softmax2(indptr1, logits1, indptr2, logits2)
+    return ref.segment_softmax2(indptr1, logits1, indptr2, logits2)
+
+
+def segment_softmax2_bwd_raw(indptr1, a1, g1, indptr2, a2, g2):
+    if use_hip(a1):
+        return get_ext().segment_softmax2_backward(indptr1, a1, g1,
+                                                   indptr2, a2, g2)
+    return ref.segment_softmax2_backward(indptr1, a1, g1, indptr2, a2, g2)
+
+
+class _SegmentSoftmax2(Function):
+    """Union softmax over two per-row edge sets (split GAT block)."""
+
+    @staticmethod
+    def forward(ctx, logits1, logits2, indptr1, indptr2):
+        a1, a2 = segment_softmax2_raw(indptr1, logits1, indptr2, logits2)
+        ctx.save_for_backward(a1, a2, indptr1, indptr2)
+        return a1, a2
+
+    @staticmethod
+    def backward(ctx, g1, g2):
+        a1, a2, indptr1, indptr2 = ctx.saved_tensors
+        d1, d2 = segment_softmax2_bwd_raw(indptr1, a1, g1.contiguous(),
+                                          indptr2, a2, g2.contiguous())
+        return d1, d2, None, None
+
+
+def segment_softmax2(logits1, logits2, indptr1, indptr2):
+    return _SegmentSoftmax2.apply(logits1, logits2, indptr1, indptr2)

@@ -436,6 +436,68 @@ __global__ void segment_softmax_bwd_kernel(const int64_t* __restrict__ indptr,
   }
 }
 
+// Union softmax over TWO per-row edge segments (split GAT block:
+// static inner edges + per-epoch sampled halo edges) — shared max/sum.
+__global__ void segment_softmax2_kernel(const int64_t* __restrict__ ip1,
+                                        const float* __restrict__ l1,
+                                        const int64_t* __restrict__ ip2,
+                                        const float* __restrict__ l2,
+                                        float* __restrict__ a1,
+                                        float* __restrict__ a2,
+                                        int n_rows, int H) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int rh = wave; rh < n_rows * H; rh += n_waves) {
+    const int r = rh / H, h = rh % H;
+    const int64_t b1 = ip1[r], e1 = ip1[r + 1];
+    const int64_t b2 = ip2[r], e2 = ip2[r + 1];
+    if (b1 == e1 && b2 == e2) continue;
+    float m = -INFINITY;
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) m = fmaxf(m, l1[e * H + h]);
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) m = fmaxf(m, l2[e * H + h]);
+    m = wave_reduce_max(m);
+    float sum = 0.f;
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) sum += __expf(l1[e * H + h] - m);
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) sum += __expf(l2[e * H + h] - m);
+    sum = wave_reduce_sum(sum);
+    const float inv = 1.0f / fmaxf(sum, 1e-38f);
+    for (int64_t e = b1 + lane; e < e1; e += WAVE)
+      a1[e * H + h] = __expf(l1[e * H + h] - m) * inv;
+    for (int64_t e = b2 + lane; e < e2; e += WAVE)
+      a2[e * H + h] = __expf(l2[e * H + h] - m) * inv;
+  }
+}
+
+__global__ void segment_softmax2_bwd_kernel(const int64_t* __restrict__ ip1,
+                                            const float* __restrict__ a1,
+                                            const float* __restrict__ g1,
+                                            const int64_t* __restrict__ ip2,
+                                            const float* __restrict__ a2,
+                                            const float* __restrict__ g2,
+                                            float* __restrict__ d1,
+                                            float* __restrict__ d2,
+                                            int n_rows, int H) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  for (int rh = wave; rh < n_rows * H; rh += n_waves) {
+    const int r = rh / H, h = rh % H;
+    const int64_t b1 = ip1[r], e1 = ip1[r + 1];
+    const int64_t b2 = ip2[r], e2 = ip2[r + 1];
+    float sum = 0.f;
+    for (int64_t e = b1 + lane; e < e1; e += WAVE)
+      sum += a1[e * H + h] * g1[e * H + h];
+    for (int64_t e = b2 + lane; e < e2; e += WAVE)
+      sum += a2[e * H + h] * g2[e * H + h];
+    sum = wave_reduce_sum(sum);
+    for (int64_t e = b1 + lane; e < e1; e += WAVE)
+      d1[e * H + h] = a1[e * H + h] * (g1[e * H + h] - sum);
+    for (int64_t e = b2 + lane; e < e2; e += WAVE)
+      d2[e * H + h] = a2[e * H + h] * (g2[e * H + h] - sum);
+  }
+}
+
 // out[r,h,:] (+)= sum_e w[e,h] * x[col_e,h,:] — wave per (row,head),
 // lanes stride D (coalesced within a head row).
 template <bool ACC>
@@ -840,6 +902,41 @@ at::Tensor segment_softmax_backward(at::Tensor indptr, at::Tensor alpha,
   return out;
 }
 
+std::vector<at::Tensor> segment_softmax2(at::Tensor ip1, at::Tensor l1,
+                                         at::Tensor ip2, at::Tensor l2) {
+  check_f32(l1, "l1"); check_f32(l2, "l2");
+  const int n_rows = ip1.numel() - 1;
+  const int H = l1.size(1);
+  auto a1 = at::empty_like(l1);
+  auto a2 = at::empty_like(l2);
+  if (n_rows == 0) return {a1, a2};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(segment_softmax2_kernel, dim3(spmm_grid(n_rows * H)),
+                     dim3(256), 0, stream, ip1.data_ptr<int64_t>(),
+                     l1.data_ptr<float>(), ip2.data_ptr<int64_t>(),
+                     l2.data_ptr<float>(), a1.data_ptr<float>(),
+                     a2.data_ptr<float>(), n_rows, H);
+  return {a1, a2};
+}
+
+std::vector<at::Tensor> segment_softmax2_backward(at::Tensor ip1, at::Tensor a1,
+                                                  at::Tensor g1, at::Tensor ip2,
+                                                  at::Tensor a2, at::Tensor g2) {
+  const int n_rows = ip1.numel() - 1;
+  const int H = a1.size(1);
+  auto d1 = at::empty_like(a1);
+  auto d2 = at::empty_like(a2);
+  if (n_rows == 0) return {d1, d2};
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(segment_softmax2_bwd_kernel, dim3(spmm_grid(n_rows * H)),
+                     dim3(256), 0, stream, ip1.data_ptr<int64_t>(),
+                     a1.data_ptr<float>(), g1.data_ptr<float>(),
+                     ip2.data_ptr<int64_t>(), a2.data_ptr<float>(),
+                     g2.data_ptr<float>(), d1.data_ptr<float>(),
+                     d2.data_ptr<float>(), n_rows, H);
+  return {d1, d2};
+}
+
 at::Tensor spmm_edge_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                          at::Tensor wstart, at::Tensor indptr,
                          at::Tensor indices, at::Tensor w,
@@ -995,6 +1092,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_softmax", &segment_softmax, "edge softmax by dst segment");
   m.def("segment_softmax_backward", &segment_softmax_backward,
         "edge softmax backward");
+  m.def("segment_softmax2", &segment_softmax2,
+        "union softmax over two edge sets (split GAT block)");
+  m.def("segment_softmax2_backward", &segment_softmax2_backward,
+        "union softmax backward");
   m.def("spmm_edge_sum", &spmm_edge_sum, "multi-head edge-weighted SpMM");
   m.def("sddmm_dot", &sddmm_dot, "per-edge per-head dot (spmm_edge grad)");
   m.def("bincount_i32", &bincount_i32, "atomic int32 histogram");

@@ -124,3 +124,49 @@ def scatter_add_rows(out: torch.Tensor, idx: torch.Tensor, src: torch.Tensor,
     s = src if scale is None else src * scale.unsqueeze(1)
     out.index_add_(0, idx.long(), s)
     return out
+
+
+def segment_softmax2(indptr1, logits1, indptr2, logits2):
+    """Softmax over the UNION of two per-row edge segments (the split GAT
+    block: static inner edges + per-epoch sampled halo edges), numerically
+    stable via the shared running max. Returns (alpha1, alpha2)."""
+    n_rows = indptr1.numel() - 1
+    H = logits1.shape[1]
+    dev, dt = logits1.device, logits1.dtype
+    row1 = _row_of_edge(indptr1)
+    row2 = _row_of_edge(indptr2)
+    m = torch.full((n_rows, H), float("-inf"), dtype=dt, device=dev)
+    if logits1.numel():
+        m = m.index_reduce_(0, row1, logits1, "amax", include_self=True)
+    if logits2.numel():
+        m = m.index_reduce_(0, row2, logits2, "amax", include_self=True)
+    m = torch.where(torch.isfinite(m), m, torch.zeros_like(m))
+    ex1 = torch.exp(logits1 - m[row1]) if logits1.numel() else logits1
+    ex2 = torch.exp(logits2 - m[row2]) if logits2.numel() else logits2
+    s = torch.zeros(n_rows, H, dtype=dt, device=dev)
+    if logits1.numel():
+        s.index_add_(0, row1, ex1)
+    if logits2.numel():
+        s.index_add_(0, row2, ex2)
+    s = s.clamp_min(1e-38)
+    a1 = ex1 / s[row1] if logits1.numel() else ex1
+    a2 = ex2 / s[row2] if logits2.numel() else ex2
+    return a1, a2
+
+
+def segment_softmax2_backward(indptr1, alpha1, grad1, indptr2, alpha2, grad2):
+    """d logits for the union softmax: dl_e = a_e * (g_e - S[row_e]) with
+    S[r] = sum over BOTH segments of a*g."""
+    n_rows = indptr1.numel() - 1
+    H = alpha1.shape[1]
+    dev, dt = alpha1.device, alpha1.dtype
+    row1 = _row_of_edge(indptr1)
+    row2 = _row_of_edge(indptr2)
+    s = torch.zeros(n_rows, H, dtype=dt, device=dev)
+    if alpha1.numel():
+        s.index_add_(0, row1, alpha1 * grad1)
+    if alpha2.numel():
+        s.index_add_(0, row2, alpha2 * grad2)
+    d1 = alpha1 * (grad1 - s[row1]) if alpha1.numel() else alpha1
+    d2 = alpha2 * (grad2 - s[row2]) if alpha2.numel() else alpha2
+    return d1, d2

@@ -42,6 +42,9 @@ class EpochState:
     # bwd: rows = recv-row index, cols = inner dst
     halo_bwd_indptr: torch.Tensor = None
     halo_bwd_indices: torch.Tensor = None
+    # weighted-transpose edge map: halo_eperm_t[j] = fwd-edge id of the
+    # j-th bwd edge (for per-edge payloads through the transpose)
+    halo_eperm_t: torch.Tensor = None
     halo_out_norm_inv: torch.Tensor = None   # [R] 1/sqrt(out_deg) of recv rows
     halo_in_deg: torch.Tensor = None         # [R] full in-degree of recv rows
 
@@ -134,7 +137,10 @@ class HaloPlan:
                 else torch.zeros(0, dtype=torch.long, device=dev))
 
         bwd_ip, bwd_ix = gather_rows_csr(self.halo_indptr, self.halo_indices, hsel)
-        fwd_ip, fwd_ix, _ = transpose_csr(bwd_ip, bwd_ix, self.n_inner)
+        fwd_ip, fwd_ix, eperm_f = transpose_csr(bwd_ip, bwd_ix, self.n_inner)
+        # eperm_f[k] = bwd edge of fwd edge k; invert for bwd -> fwd
+        eperm_t = torch.empty_like(eperm_f)
+        eperm_t[eperm_f] = torch.arange(eperm_f.numel(), device=eperm_f.device)
         if dev.type == "cuda":
             from ..ops.functional import _worklist_of
             _worklist_of(bwd_ip)
@@ -146,6 +152,7 @@ class HaloPlan:
             pack_idx=pack_idx, pack_scale=pack_scale, hsel=hsel,
             halo_fwd_indptr=fwd_ip, halo_fwd_indices=fwd_ix,
             halo_bwd_indptr=bwd_ip, halo_bwd_indices=bwd_ix,
+            halo_eperm_t=eperm_t,
             halo_out_norm_inv=self.halo_out_norm_inv_full[hsel],
             halo_in_deg=self.halo_in_deg_full[hsel],
         )

@@ -341,3 +341,24 @@ def test_train_and_eval_on_gpu_accuracy():
     final.load_state_dict({k: v.cpu() for k, v in m.state_dict().items()})
     out = ev.evaluate(final)
     assert out["test"] > 0.30, out
+
+
+@needs_gpu
+def test_segment_softmax2_matches_reference():
+    from bnsgcn_amd.ops.functional import segment_softmax2_raw
+    H = 4
+    ip1, _ = rand_csr(300, 10, 4000, seed=41)
+    ip2, _ = rand_csr(300, 10, 700, seed=42)
+    l1 = torch.randn(4000, H)
+    l2 = torch.randn(700, H)
+    w1, w2 = ref.segment_softmax2(ip1, l1, ip2, l2)
+    g1, g2 = segment_softmax2_raw(ip1.cuda(), l1.cuda(), ip2.cuda(), l2.cuda())
+    torch.testing.assert_close(g1.cpu(), w1, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(g2.cpu(), w2, rtol=1e-4, atol=1e-5)
+    ga, gb = torch.randn_like(l1), torch.randn_like(l2)
+    wd1, wd2 = ref.segment_softmax2_backward(ip1, w1, ga, ip2, w2, gb)
+    from bnsgcn_amd.ops.functional import segment_softmax2_bwd_raw
+    gd1, gd2 = segment_softmax2_bwd_raw(ip1.cuda(), g1, ga.cuda(),
+                                        ip2.cuda(), g2, gb.cuda())
+    torch.testing.assert_close(gd1.cpu(), wd1, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(gd2.cpu(), wd2, rtol=1e-4, atol=1e-5)

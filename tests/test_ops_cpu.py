@@ -278,3 +278,57 @@ def test_gradcheck_custom_functions_f64():
                                               indices_t, eperm_t), (el, er))
     lg = torch.randn(30, H, dtype=torch.float64, requires_grad=True)
     assert gradcheck(lambda t: F.segment_softmax(t, indptr), (lg,))
+
+
+def test_segment_softmax2_matches_combined():
+    """Union softmax over two edge sets == softmax over the merged set."""
+    from bnsgcn_amd.ops.csr_torch import merge_csr
+    H = 3
+    ip1, ix1 = rand_csr(12, 9, 50, seed=31)
+    ip2, ix2 = rand_csr(12, 5, 20, seed=32)
+    l1 = torch.randn(50, H, requires_grad=True)
+    l2 = torch.randn(20, H, requires_grad=True)
+    a1, a2 = F.segment_softmax2(l1, l2, ip1, ip2)
+    # oracle: merged CSR + plain segment softmax; set-1 edges precede set-2
+    mip, _ = merge_csr(ip1, ix1, ip2, ix2, col_offset2=9)
+    l1b = l1.detach().clone().requires_grad_(True)
+    l2b = l2.detach().clone().requires_grad_(True)
+    lm = torch.empty(70, H)
+    n1 = ip1[1:] - ip1[:-1]
+    n2 = ip2[1:] - ip2[:-1]
+    pos1, pos2 = [], []
+    o = 0
+    for r in range(12):
+        pos1 += list(range(o, o + int(n1[r])))
+        o += int(n1[r])
+        pos2 += list(range(o, o + int(n2[r])))
+        o += int(n2[r])
+    idx1 = torch.tensor(pos1)
+    idx2 = torch.tensor(pos2)
+    lm = torch.zeros(70, H)
+    lm[idx1] = l1b
+    lm[idx2] = l2b
+    am = ref.segment_softmax(mip, lm)
+    torch.testing.assert_close(a1, am[idx1], rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(a2, am[idx2], rtol=1e-5, atol=1e-6)
+    # gradients agree with autograd through the merged formulation
+    g1 = torch.randn_like(a1)
+    g2 = torch.randn_like(a2)
+    (a1 * g1).sum().backward(retain_graph=True)
+    (a2 * g2).sum().backward()
+    amf = F.segment_softmax(lm, mip)
+    gm = torch.zeros_like(amf)
+    gm[idx1] = g1
+    gm[idx2] = g2
+    (amf * gm).sum().backward()
+    torch.testing.assert_close(l1.grad, l1b.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(l2.grad, l2b.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_segment_softmax2_gradcheck():
+    from torch.autograd import gradcheck
+    ip1, _ = rand_csr(6, 5, 18, seed=33)
+    ip2, _ = rand_csr(6, 4, 9, seed=34)
+    l1 = torch.randn(18, 2, dtype=torch.float64, requires_grad=True)
+    l2 = torch.randn(9, 2, dtype=torch.float64, requires_grad=True)
+    assert gradcheck(lambda a, b: F.segment_softmax2(a, b, ip1, ip2), (l1, l2))
