@@ -1028,6 +1028,12 @@ at::Tensor gemm_strided(const at::Tensor& A, int64_t sAm, int64_t sAk,
                         const c10::optional<at::Tensor>& bias,
                         int M, int N, int K) {
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (M == 0 || N == 0) return at::zeros({M, N}, A.options());
+  if (K == 0) {
+    auto C = at::zeros({M, N}, A.options());
+    if (bias.has_value()) C += *bias;
+    return C;
+  }
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
   // split-K when the M/N tile grid alone cannot fill the 256 CUs (dW
   // reductions: C is tiny, K is the node count)

@@ -362,3 +362,18 @@ def test_segment_softmax2_matches_reference():
                                         ip2.cuda(), g2, gb.cuda())
     torch.testing.assert_close(gd1.cpu(), wd1, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(gd2.cpu(), wd2, rtol=1e-4, atol=1e-5)
+
+
+@needs_gpu
+def test_gemm_empty_inputs():
+    """0-row / 0-col GEMMs must not crash (empty-halo GAT at world size 1
+    hit a divide-by-zero in split-K block sizing)."""
+    x0 = torch.zeros(0, 300).cuda()
+    w = torch.randn(64, 300).cuda()
+    b = torch.randn(64).cuda()
+    out = ext.gemm_nt_bias(x0, w, b)
+    assert out.shape == (0, 64)
+    g0 = torch.zeros(0, 64).cuda()
+    assert ext.gemm_nn(g0, w).shape == (0, 300)
+    dw = ext.gemm_tn(g0, x0)
+    torch.testing.assert_close(dw, torch.zeros(64, 300).cuda())
