@@ -58,8 +58,9 @@ def build_worklist(indptr: torch.Tensor, seg: int = None, max_waves: int = None
       contiguous slab of the row space for its private L2.
 
     Returns (wrow int32 [W_items] (negative ~row = atomic), wbeg, wend
-    int64, wave_start int32 [n_waves+1]); n_waves is a multiple of 4
-    (one 256-thread block = 4 waves).
+    int64, wave_start int32 [n_waves+1]); n_waves is a multiple of 8
+    (one 256-thread block = 4 waves of 64 lanes, or 8 half-wave subgroups
+    for the narrow-F kernel variant).
     """
     seg = seg or SEG
     max_waves = max_waves or MAX_WAVES
@@ -91,8 +92,8 @@ def build_worklist(indptr: torch.Tensor, seg: int = None, max_waves: int = None
     lens = (wend - wbeg)
     cum = torch.cumsum(torch.cat([torch.zeros(1, dtype=lens.dtype, device=device),
                                   lens]), 0)
-    n_waves = min(max_waves, max(4, total))
-    n_waves = (n_waves + 3) // 4 * 4
+    n_waves = min(max_waves, max(8, total))
+    n_waves = (n_waves + 7) // 8 * 8
     targets = (cum[-1] * torch.arange(n_waves + 1, device=device).double()
                / n_waves).to(cum.dtype)
     wave_start = torch.searchsorted(cum, targets).to(torch.int32)

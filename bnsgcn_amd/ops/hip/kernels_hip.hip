@@ -42,16 +42,19 @@ DEV_INLINE int xcd_remap_block(int b, int nb) {
   return (xcd < r) ? xcd * (q + 1) + j : r * (q + 1) + (xcd - r) * q + j;
 }
 
-template <bool ACC>
+template <bool ACC, int SUBW>
 __global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
     const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
     const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
     const int32_t* __restrict__ indices, const float* __restrict__ x,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
     float* __restrict__ out, int f4) {
+  // SUBW = scheduling width: 64 (full wave) for wide rows, 32 (half-wave
+  // subgroups, independent items per half) when f4 <= 32 so narrow
+  // feature dims (e.g. h=128 -> f4=32) keep every lane busy.
   const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
-  const int w = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
-  const int lane = threadIdx.x & (WAVE - 1);
+  const int w = bb * (blockDim.x / SUBW) + (threadIdx.x / SUBW);
+  const int lane = threadIdx.x & (SUBW - 1);
   const int it_beg = wave_start[w], it_end = wave_start[w + 1];
   const float4* __restrict__ x4 = reinterpret_cast<const float4*>(x);
   float4* __restrict__ out4 = reinterpret_cast<float4*>(out);
@@ -62,13 +65,13 @@ __global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
     if (atomic) row = ~row;
     const int64_t beg = wbeg[it], end = wend[it];
     const float ds = dst_scale ? dst_scale[row] : 1.0f;
-    for (int f0 = 0; f0 < f4; f0 += 2 * WAVE) {
+    for (int f0 = 0; f0 < f4; f0 += 2 * SUBW) {
       const int fA = f0 + lane;
-      const int fB = fA + WAVE;
+      const int fB = fA + SUBW;
       const bool hasA = fA < f4, hasB = fB < f4;
       float4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
-      for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
-        const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
+      for (int64_t e0 = beg; e0 < end; e0 += SUBW) {
+        const int nv = (int)((end - e0 < SUBW) ? (end - e0) : SUBW);
         int cid = 0;
         float ssc = 1.0f;
         if (lane < nv) {
@@ -77,8 +80,8 @@ __global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
         }
 #pragma unroll 4
         for (int k = 0; k < nv; ++k) {
-          const int c = __shfl(cid, k, WAVE);
-          const float ss = src_scale ? __shfl(ssc, k, WAVE) : 1.0f;
+          const int c = __shfl(cid, k, SUBW);
+          const float ss = src_scale ? __shfl(ssc, k, SUBW) : 1.0f;
           const int64_t base = (int64_t)c * f4;
           if (hasA) f4_axpy(acc0, ss, x4[base + fA]);
           if (hasB) f4_axpy(acc1, ss, x4[base + fB]);
@@ -769,7 +772,19 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int grid = n_waves / 4;
   if (F % 4 == 0) {
-    auto kfn = acc ? spmm_sum_vec4_kernel<true> : spmm_sum_vec4_kernel<false>;
+    if (F / 4 <= 32 && n_waves % 8 == 0) {
+      auto kfn = acc ? spmm_sum_vec4_kernel<true, 32>
+                     : spmm_sum_vec4_kernel<false, 32>;
+      hipLaunchKernelGGL(kfn, dim3(n_waves / 8), dim3(256), 0, stream,
+                         wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
+                         wend.data_ptr<int64_t>(), wave_start.data_ptr<int32_t>(),
+                         indices.data_ptr<int32_t>(), x.data_ptr<float>(),
+                         opt_ptr(src_scale), opt_ptr(dst_scale),
+                         out.data_ptr<float>(), F / 4);
+      return out;
+    }
+    auto kfn = acc ? spmm_sum_vec4_kernel<true, 64>
+                   : spmm_sum_vec4_kernel<false, 64>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
                        wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
                        wend.data_ptr<int64_t>(), wave_start.data_ptr<int32_t>(),
