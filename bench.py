@@ -85,6 +85,15 @@ def main():
     args.graph_name = graph_name_of(args)
 
     cuda = torch.cuda.is_available() and a.device != "cpu"
+    # set the device BEFORE the first collective: an NCCL barrier issued
+    # while every rank still sits on cuda:0 builds duplicate-device
+    # communicators and aborts the whole 8-GPU job
+    if a.device == "auto":
+        device = f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}" if cuda else "cpu"
+    else:
+        device = a.device
+    if cuda:
+        torch.cuda.set_device(torch.device(device))
     if world > 1:
         rank, world = init_distributed("nccl" if cuda else "gloo")
     if rank == 0:
@@ -92,13 +101,6 @@ def main():
         prepare_partitions(args)
     if world > 1:
         dist.barrier()
-
-    if a.device == "auto":
-        device = f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}" if cuda else "cpu"
-    else:
-        device = a.device
-    if cuda:
-        torch.cuda.set_device(torch.device(device))
 
     part = load_partition(args.partition_dir, args.graph_name, rank)
     meta = part.meta
