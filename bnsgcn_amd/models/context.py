@@ -11,12 +11,10 @@ tensors; the per-epoch part lives in HaloPlan.set_epoch.
 """
 from __future__ import annotations
 
-import numpy as np
 import torch
 
 from ..graph.store import Partition
 from ..ops.csr_torch import transpose_csr, merge_csr
-from ..ops.functional import spmm_sum_raw
 from ..parallel.halo import partition_aggregate
 from ..parallel.plan import EpochState, HaloPlan
 

@@ -10,7 +10,6 @@ the reference so checkpoints stay structurally compatible.
 """
 from __future__ import annotations
 
-import torch
 from torch import nn
 
 from .context import GraphContext
