@@ -34,50 +34,55 @@ def _uniform_init(*tensors, fan_in):
 
 
 class GCNLayer(nn.Module):
+    """State-dict keys match the reference exactly (layers.<i>.linear.*,
+    module/layer.py:17) so checkpoints are interchangeable."""
+
     def __init__(self, in_feats, out_feats, bias=True, use_pp=False):
         super().__init__()
         self.use_pp = use_pp
-        self.weight = nn.Parameter(torch.empty(out_feats, in_feats))
-        self.bias = nn.Parameter(torch.empty(out_feats)) if bias else None
-        _uniform_init(self.weight, fan_in=in_feats)
-        if self.bias is not None:
-            _uniform_init(self.bias, fan_in=in_feats)
+        self.linear = nn.Linear(in_feats, out_feats, bias=bias)
+        _uniform_init(self.linear.weight, fan_in=in_feats)
+        if self.linear.bias is not None:
+            _uniform_init(self.linear.bias, fan_in=in_feats)
 
     def forward(self, ctx: GraphContext, x):
         if self.training and self.use_pp:
-            return F.linear(x, self.weight, self.bias)
+            return F.linear(x, self.linear.weight, self.linear.bias)
         h = ctx.aggregate(x, "gcn")
-        return F.linear(h, self.weight, self.bias)
+        return F.linear(h, self.linear.weight, self.linear.bias)
 
 
 class SAGELayer(nn.Module):
+    """State-dict keys match the reference (layers.<i>.linear.* under
+    use_pp, layers.<i>.linear{1,2}.* otherwise — module/layer.py:59-62)."""
+
     def __init__(self, in_feats, out_feats, bias=True, use_pp=False):
         super().__init__()
         self.use_pp = use_pp
         if use_pp:
-            self.weight = nn.Parameter(torch.empty(out_feats, 2 * in_feats))
-            self.bias = nn.Parameter(torch.empty(out_feats)) if bias else None
-            _uniform_init(self.weight, fan_in=2 * in_feats)
-            if self.bias is not None:
-                _uniform_init(self.bias, fan_in=2 * in_feats)
+            self.linear = nn.Linear(2 * in_feats, out_feats, bias=bias)
+            _uniform_init(self.linear.weight, fan_in=2 * in_feats)
+            if self.linear.bias is not None:
+                _uniform_init(self.linear.bias, fan_in=2 * in_feats)
         else:
-            self.weight1 = nn.Parameter(torch.empty(out_feats, in_feats))
-            self.weight2 = nn.Parameter(torch.empty(out_feats, in_feats))
-            self.bias1 = nn.Parameter(torch.empty(out_feats)) if bias else None
-            self.bias2 = nn.Parameter(torch.empty(out_feats)) if bias else None
-            _uniform_init(self.weight1, self.weight2, fan_in=in_feats)
+            self.linear1 = nn.Linear(in_feats, out_feats, bias=bias)
+            self.linear2 = nn.Linear(in_feats, out_feats, bias=bias)
+            _uniform_init(self.linear1.weight, self.linear2.weight,
+                          fan_in=in_feats)
             if bias:
-                _uniform_init(self.bias1, self.bias2, fan_in=in_feats)
+                _uniform_init(self.linear1.bias, self.linear2.bias,
+                              fan_in=in_feats)
 
     def forward(self, ctx: GraphContext, x):
         if self.training and self.use_pp:
             # x = [feat ‖ precomputed neighbor mean], width 2F
-            return F.linear(x, self.weight, self.bias)
+            return F.linear(x, self.linear.weight, self.linear.bias)
         ah = ctx.aggregate(x, "mean")
         if self.use_pp:  # eval path of a pp layer (reference layer.py:98-100)
-            return F.linear(torch.cat((x, ah), dim=1), self.weight, self.bias)
-        return (F.linear(x, self.weight1, self.bias1)
-                + F.linear(ah, self.weight2, self.bias2))
+            return F.linear(torch.cat((x, ah), dim=1), self.linear.weight,
+                            self.linear.bias)
+        return (F.linear(x, self.linear1.weight, self.linear1.bias)
+                + F.linear(ah, self.linear2.weight, self.linear2.bias))
 
 
 class GATLayer(nn.Module):
@@ -86,7 +91,9 @@ class GATLayer(nn.Module):
         super().__init__()
         self.heads, self.out_feats = heads, out_feats
         self.use_pp = use_pp
-        self.fc = nn.Parameter(torch.empty(heads * out_feats, in_feats))
+        # param names follow DGL GATConv (fc.weight, attn_l, attn_r, bias)
+        # so reference GAT checkpoints load directly
+        self.fc = nn.Linear(in_feats, heads * out_feats, bias=False)
         self.attn_l = nn.Parameter(torch.empty(1, heads, out_feats))
         self.attn_r = nn.Parameter(torch.empty(1, heads, out_feats))
         self.bias = nn.Parameter(torch.zeros(heads * out_feats)) if bias else None
@@ -94,7 +101,7 @@ class GATLayer(nn.Module):
         self.attn_drop = nn.Dropout(attn_drop)
         self.negative_slope = negative_slope
         gain = math.sqrt(2.0)
-        nn.init.xavier_normal_(self.fc, gain=gain)
+        nn.init.xavier_normal_(self.fc.weight, gain=gain)
         nn.init.xavier_normal_(self.attn_l, gain=gain)
         nn.init.xavier_normal_(self.attn_r, gain=gain)
 
@@ -116,7 +123,7 @@ class GATLayer(nn.Module):
         H, D = self.heads, self.out_feats
         if ctx.plan is None:                       # full-graph eval path
             ip, ix, tip, tix, eperm, _ = ctx.gat_block()
-            z = F.linear(h, self.fc).view(-1, H, D)
+            z = F.linear(h, self.fc.weight).view(-1, H, D)
             el = (z * self.attn_l).sum(-1)
             er = (z * self.attn_r).sum(-1)
             logits = F.sddmm_add(el, er, ip, ix, tip, tix, eperm)
@@ -132,8 +139,8 @@ class GATLayer(nn.Module):
             src_halo = halo_exchange(h, ctx.plan)
             hip, hix, hbip, hbix, heperm = ctx.gat_split_halo()
 
-        z_in = F.linear(h, self.fc).view(-1, H, D)
-        z_h = F.linear(src_halo, self.fc).view(-1, H, D)
+        z_in = F.linear(h, self.fc.weight).view(-1, H, D)
+        z_h = F.linear(src_halo, self.fc.weight).view(-1, H, D)
         el_in = (z_in * self.attn_l).sum(-1)       # inner sources
         el_h = (z_h * self.attn_l).sum(-1)         # halo sources
         er = (z_in * self.attn_r).sum(-1)          # inner destinations

@@ -171,3 +171,29 @@ def test_resume_from_checkpoint(tmp_path):
         assert np.isfinite(res[0]["loss_history"]).all()
     finally:
         os.chdir(cwd)
+
+
+def test_state_dict_keys_match_reference_naming():
+    """Checkpoint key layout matches the reference models (module/layer.py
+    nn.Linear submodules; DGL GATConv names for GAT) so checkpoints are
+    interchangeable."""
+    from bnsgcn_amd.models.models import create_model
+    from pathlib import Path
+    tmp = Path("/tmp")
+    args = make_args(tmp, model="graphsage", use_pp=True, n_layers=3,
+                     n_linear=1)
+    m = create_model(args, n_feat=16, n_class=7, train_size=100)
+    keys = set(m.state_dict().keys())
+    assert "layers.0.linear.weight" in keys     # pp SAGE layer
+    assert "layers.1.linear1.weight" in keys    # plain SAGE layer
+    assert "layers.1.linear2.bias" in keys
+    assert "layers.2.weight" in keys            # nn.Linear MLP tail
+    assert "norm.0.weight" in keys              # LayerNorm
+    args = make_args(tmp, model="gcn", use_pp=False, n_layers=2)
+    m = create_model(args, n_feat=16, n_class=7, train_size=100)
+    assert "layers.0.linear.weight" in m.state_dict()
+    args = make_args(tmp, model="gat", heads=2, n_layers=2)
+    m = create_model(args, n_feat=16, n_class=7, train_size=100)
+    keys = set(m.state_dict().keys())
+    assert {"layers.0.fc.weight", "layers.0.attn_l", "layers.0.attn_r",
+            "layers.0.bias"} <= keys
