@@ -126,8 +126,8 @@ class GATLayer(nn.Module):
             z = F.linear(h, self.fc.weight).view(-1, H, D)
             el = (z * self.attn_l).sum(-1)
             er = (z * self.attn_r).sum(-1)
-            logits = F.sddmm_add(el, er, ip, ix, tip, tix, eperm)
-            logits = torch.nn.functional.leaky_relu(logits, self.negative_slope)
+            logits = F.sddmm_add(el, er, ip, ix, tip, tix, eperm,
+                                 slope=self.negative_slope)
             alpha = self.attn_drop(F.segment_softmax(logits, ip))
             out = F.spmm_edge_sum(z, alpha, ip, ix, tip, tix, eperm)
             return out + self.bias.view(1, H, D) if self.bias is not None else out
@@ -144,12 +144,10 @@ class GATLayer(nn.Module):
         el_in = (z_in * self.attn_l).sum(-1)       # inner sources
         el_h = (z_h * self.attn_l).sum(-1)         # halo sources
         er = (z_in * self.attn_r).sum(-1)          # inner destinations
-        li = F.sddmm_add(el_in, er, ctx.indptr, ctx.indices,
-                         ctx.t_indptr, ctx.t_indices, ctx.t_eperm)
-        lh = F.sddmm_add(el_h, er, hip, hix, hbip, hbix, heperm)
         slope = self.negative_slope
-        li = torch.nn.functional.leaky_relu(li, slope)
-        lh = torch.nn.functional.leaky_relu(lh, slope)
+        li = F.sddmm_add(el_in, er, ctx.indptr, ctx.indices,
+                         ctx.t_indptr, ctx.t_indices, ctx.t_eperm, slope=slope)
+        lh = F.sddmm_add(el_h, er, hip, hix, hbip, hbix, heperm, slope=slope)
         ai, ah = F.segment_softmax2(li, lh, ctx.indptr, hip)
         ai = self.attn_drop(ai)
         ah = self.attn_drop(ah)

@@ -56,10 +56,18 @@ def sddmm_dot_raw(indptr, indices, a_dst, b_src):
     return ref.sddmm_dot(indptr, indices, a_dst, b_src)
 
 
-def sddmm_add_raw(indptr, indices, el_src, er_dst):
+def sddmm_add_raw(indptr, indices, el_src, er_dst, slope=-1.0):
+    """slope >= 0 fuses LeakyReLU into the kernel (HIP path); the torch
+    path applies it separately."""
     if use_hip(el_src):
-        return get_ext().sddmm_add(indptr, indices, el_src, er_dst)
-    return ref.sddmm_add(indptr, indices, el_src, er_dst)
+        wl = _worklist_of(indptr)
+        return get_ext().sddmm_add(*wl, indptr, indices,
+                                   el_src.contiguous(), er_dst.contiguous(),
+                                   slope)
+    out = ref.sddmm_add(indptr, indices, el_src, er_dst)
+    if slope >= 0:
+        out = torch.nn.functional.leaky_relu(out, slope)
+    return out
 
 
 def segment_softmax_raw(indptr, logits):
@@ -140,18 +148,34 @@ def spmm_edge_sum(x, w, indptr, indices, indptr_t, indices_t, eperm_t):
 
 
 class _SDDMMAdd(Function):
-    """logits[e,h] = el[col_e,h] + er[row_e,h]  (u_add_v SDDMM)."""
+    """logits[e,h] = [leaky_relu](el[col_e,h] + er[row_e,h])  (u_add_v
+    SDDMM, optionally fused with the GAT LeakyReLU — the activation is
+    monotonic with fixpoint 0, so the backward mask comes from the sign
+    of the saved OUTPUT)."""
 
     @staticmethod
-    def forward(ctx, el, er, indptr, indices, indptr_t, indices_t, eperm_t):
-        ctx.save_for_backward(indptr, indices, indptr_t, indices_t, eperm_t)
+    def forward(ctx, el, er, indptr, indices, indptr_t, indices_t, eperm_t,
+                slope):
+        out = sddmm_add_raw(indptr, indices, el, er,
+                            -1.0 if slope is None else float(slope))
+        ctx.slope = slope
+        if slope is not None:
+            ctx.save_for_backward(indptr, indices, indptr_t, indices_t,
+                                  eperm_t, out)
+        else:
+            ctx.save_for_backward(indptr, indices, indptr_t, indices_t,
+                                  eperm_t)
         ctx.n_src, ctx.n_dst = el.shape[0], er.shape[0]
-        return sddmm_add_raw(indptr, indices, el, er)
+        return out
 
     @staticmethod
     def backward(ctx, grad):
-        indptr, indices, indptr_t, indices_t, eperm_t = ctx.saved_tensors
-        grad = grad.contiguous()
+        if ctx.slope is not None:
+            indptr, indices, indptr_t, indices_t, eperm_t, out = ctx.saved_tensors
+            grad = torch.where(out > 0, grad, grad * ctx.slope).contiguous()
+        else:
+            indptr, indices, indptr_t, indices_t, eperm_t = ctx.saved_tensors
+            grad = grad.contiguous()
         g_el = g_er = None
         if use_hip(grad) and grad.shape[1] <= 8:
             e = get_ext()
@@ -163,7 +187,7 @@ class _SDDMMAdd(Function):
             if ctx.needs_input_grad[1]:
                 g_er = e.segment_sum_edges(*_worklist_of(indptr), None,
                                            grad, ctx.n_dst)
-            return g_el, g_er, None, None, None, None, None
+            return g_el, g_er, None, None, None, None, None, None
         if ctx.needs_input_grad[0]:
             g_el = torch.zeros(ctx.n_src, grad.shape[1], dtype=grad.dtype,
                                device=grad.device)
@@ -175,11 +199,13 @@ class _SDDMMAdd(Function):
                 torch.arange(ctx.n_dst, device=grad.device),
                 indptr[1:] - indptr[:-1])
             g_er.index_add_(0, row, grad)
-        return g_el, g_er, None, None, None, None, None
+        return g_el, g_er, None, None, None, None, None, None
 
 
-def sddmm_add(el, er, indptr, indices, indptr_t, indices_t, eperm_t):
-    return _SDDMMAdd.apply(el, er, indptr, indices, indptr_t, indices_t, eperm_t)
+def sddmm_add(el, er, indptr, indices, indptr_t, indices_t, eperm_t,
+              slope=None):
+    return _SDDMMAdd.apply(el, er, indptr, indices, indptr_t, indices_t,
+                           eperm_t, slope)
 
 
 class _SegmentSoftmax(Function):

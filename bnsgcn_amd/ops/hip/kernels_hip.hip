@@ -371,23 +371,32 @@ DEV_INLINE float wave_reduce_max(float v) {
   return v;
 }
 
-// logits[e,h] = el[col_e,h] + er[row_e,h] — edge-parallel, row via the
-// contiguous CSR segment (wave per row, lanes stride edges*heads).
-__global__ void sddmm_add_kernel(const int64_t* __restrict__ indptr,
-                                 const int32_t* __restrict__ indices,
-                                 const float* __restrict__ el,
-                                 const float* __restrict__ er,
-                                 float* __restrict__ out, int n_rows, int H) {
-  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+// logits[e,h] = leaky_relu(el[col_e,h] + er[row_e,h]) — work-list
+// scheduled with batched coalesced index loads (the row-serial form was
+// dependent-load bound at ~200 GB/s); slope < 0 disables the activation.
+__global__ __launch_bounds__(256) void sddmm_add_kernel(
+    const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
+    const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
+    const int32_t* __restrict__ indices, const float* __restrict__ el,
+    const float* __restrict__ er, float* __restrict__ out, int H,
+    float slope) {
+  const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
+  const int wv = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
   const int lane = threadIdx.x & (WAVE - 1);
-  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
-  for (int r = wave; r < n_rows; r += n_waves) {
-    const int64_t beg = indptr[r], end = indptr[r + 1];
-    const int64_t cnt = (end - beg) * H;
-    for (int64_t t = lane; t < cnt; t += WAVE) {
-      const int64_t e = beg + t / H;
-      const int h = t % H;
-      out[e * H + h] = el[(int64_t)indices[e] * H + h] + er[(int64_t)r * H + h];
+  const int it_beg = wave_start[wv], it_end = wave_start[wv + 1];
+  for (int it = it_beg; it < it_end; ++it) {
+    int row = wrow[it];
+    if (row < 0) row = ~row;
+    const int64_t beg = wbeg[it], end = wend[it];
+    for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
+      const int64_t e = e0 + lane;
+      if (e >= end) break;
+      const int64_t c = indices[e];
+      for (int h = 0; h < H; ++h) {
+        float v = el[c * H + h] + er[(int64_t)row * H + h];
+        if (slope >= 0.f && v < 0.f) v *= slope;
+        out[e * H + h] = v;
+      }
     }
   }
 }
@@ -874,18 +883,21 @@ at::Tensor bincount_i32(at::Tensor v, int64_t n_bins) {
   return out;
 }
 
-at::Tensor sddmm_add(at::Tensor indptr, at::Tensor indices, at::Tensor el,
-                     at::Tensor er) {
+at::Tensor sddmm_add(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
+                     at::Tensor wstart, at::Tensor indptr, at::Tensor indices,
+                     at::Tensor el, at::Tensor er, double slope) {
   check_f32(el, "el"); check_f32(er, "er");
-  const int n_rows = indptr.numel() - 1;
   const int H = el.size(1);
   auto out = at::empty({indices.numel(), H}, el.options());
-  if (indices.numel() == 0) return out;
+  const int n_waves = wstart.numel() - 1;
+  if (indices.numel() == 0 || n_waves <= 0) return out;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(sddmm_add_kernel, dim3(spmm_grid(n_rows)), dim3(256), 0,
-                     stream, indptr.data_ptr<int64_t>(),
+  hipLaunchKernelGGL(sddmm_add_kernel, dim3(n_waves / 4), dim3(256), 0,
+                     stream, wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
+                     wend.data_ptr<int64_t>(), wstart.data_ptr<int32_t>(),
                      indices.data_ptr<int32_t>(), el.data_ptr<float>(),
-                     er.data_ptr<float>(), out.data_ptr<float>(), n_rows, H);
+                     er.data_ptr<float>(), out.data_ptr<float>(), H,
+                     (float)slope);
   return out;
 }
 
