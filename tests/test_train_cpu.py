@@ -197,3 +197,30 @@ def test_state_dict_keys_match_reference_naming():
     keys = set(m.state_dict().keys())
     assert {"layers.0.fc.weight", "layers.0.attn_l", "layers.0.attn_r",
             "layers.0.bias"} <= keys
+
+
+@pytest.mark.parametrize("seed", [0, 1])
+def test_fuzz_configs(tmp_path, seed):
+    """Randomized config fuzz: models x partition counts x rates x flags,
+    a few epochs each — everything must stay finite (shape corner cases:
+    small partitions, empty boundaries, odd hidden sizes, n_linear)."""
+    import random
+    rng = random.Random(seed)
+    for trial in range(4):
+        world = rng.choice([2, 3])
+        kw = dict(
+            model=rng.choice(["graphsage", "gcn", "gat"]),
+            sampling_rate=rng.choice([0.0, 0.15, 0.6, 1.0]),
+            use_pp=rng.choice([True, False]),
+            n_layers=rng.choice([2, 3]),
+            n_hidden=rng.choice([8, 12, 24]),
+            heads=rng.choice([1, 2]),
+            n_epochs=3,
+            dropout=rng.choice([0.0, 0.3]),
+            inductive=rng.choice([True, False]),
+            partition_method=rng.choice(["metis", "random"]),
+            n_linear=rng.choice([0, 1]),
+        )
+        res = _run_config(tmp_path / f"f{seed}_{trial}", world, **kw)
+        for m in res:
+            assert np.isfinite(m["loss_history"]).all(), (kw, m)
