@@ -59,6 +59,9 @@ def all_to_all_rows(recv: torch.Tensor, send: torch.Tensor,
     order; recv: preallocated [sum(recv_counts), F]. Returns a waitable
     handle when async_op (nccl) else None.
     """
+    if sum(send_counts) == 0 and sum(recv_counts) == 0:
+        return None          # sampling-rate 0: nothing moves (NCCL dislikes
+                             # empty tensors in collectives)
     if not dist.is_initialized() or dist.get_world_size() == 1:
         # single-process: only the (normally empty) self block
         if recv_counts and recv_counts[0] > 0:
