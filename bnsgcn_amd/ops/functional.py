@@ -30,9 +30,17 @@ def _worklist_of(indptr):
 
 def spmm_sum_raw(indptr, indices, x, src_scale=None, dst_scale=None, out=None):
     if use_hip(x):
-        wrow, wbeg, wend, wave_start = _worklist_of(indptr)
+        wrow, wbeg, wend, wave_start, zero_rows = _worklist_of(indptr)
+        acc = out is not None
+        if not acc:
+            # allocate uninitialized and zero ONLY the rows the kernel does
+            # not fully overwrite (split/atomic + empty rows)
+            out = torch.empty(indptr.numel() - 1, x.shape[1],
+                              dtype=x.dtype, device=x.device)
+            if zero_rows.numel():
+                out.index_fill_(0, zero_rows, 0.0)
         return get_ext().spmm_sum(wrow, wbeg, wend, wave_start, indices, x,
-                                  indptr.numel() - 1, src_scale, dst_scale, out)
+                                  src_scale, dst_scale, out, acc)
     return ref.spmm_sum(indptr, indices, x, src_scale, dst_scale, out)
 
 
@@ -40,7 +48,7 @@ def spmm_edge_raw(indptr, indices, eweight, x, out=None, wperm=None):
     """wperm: optional edge permutation applied to eweight INSIDE the HIP
     kernel (fuses the w[eperm] gather of the transposed backward pass)."""
     if use_hip(x):
-        wl = _worklist_of(indptr)
+        wl = _worklist_of(indptr)[:4]
         return get_ext().spmm_edge_sum(*wl, indptr, indices,
                                        eweight.contiguous(), wperm,
                                        x.contiguous(), out)
@@ -50,7 +58,7 @@ def spmm_edge_raw(indptr, indices, eweight, x, out=None, wperm=None):
 
 def sddmm_dot_raw(indptr, indices, a_dst, b_src):
     if use_hip(a_dst):
-        wl = _worklist_of(indptr)
+        wl = _worklist_of(indptr)[:4]
         return get_ext().sddmm_dot(*wl, indptr, indices,
                                    a_dst.contiguous(), b_src.contiguous())
     return ref.sddmm_dot(indptr, indices, a_dst, b_src)
@@ -60,7 +68,7 @@ def sddmm_add_raw(indptr, indices, el_src, er_dst, slope=-1.0):
     """slope >= 0 fuses LeakyReLU into the kernel (HIP path); the torch
     path applies it separately."""
     if use_hip(el_src):
-        wl = _worklist_of(indptr)
+        wl = _worklist_of(indptr)[:4]
         return get_ext().sddmm_add(*wl, indptr, indices,
                                    el_src.contiguous(), er_dst.contiguous(),
                                    slope)
@@ -182,10 +190,10 @@ class _SDDMMAdd(Function):
             if ctx.needs_input_grad[0]:
                 # d el[c] = Σ_{e: col_e=c} grad[e]: segment-sum over the
                 # transposed CSR with the edge permutation
-                g_el = e.segment_sum_edges(*_worklist_of(indptr_t), eperm_t,
+                g_el = e.segment_sum_edges(*_worklist_of(indptr_t)[:4], eperm_t,
                                            grad, ctx.n_src)
             if ctx.needs_input_grad[1]:
-                g_er = e.segment_sum_edges(*_worklist_of(indptr), None,
+                g_er = e.segment_sum_edges(*_worklist_of(indptr)[:4], None,
                                            grad, ctx.n_dst)
             return g_el, g_er, None, None, None, None, None, None
         if ctx.needs_input_grad[0]:

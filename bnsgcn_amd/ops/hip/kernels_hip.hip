@@ -761,11 +761,14 @@ int spmm_grid(int n_rows) {
 
 at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                     at::Tensor wave_start, at::Tensor indices, at::Tensor x,
-                    int64_t n_rows,
                     c10::optional<at::Tensor> src_scale,
                     c10::optional<at::Tensor> dst_scale,
-                    c10::optional<at::Tensor> out_opt) {
+                    at::Tensor out, bool acc) {
+  // `out` is always caller-allocated: accumulate when acc, otherwise the
+  // kernel overwrites every non-empty row (caller pre-zeroed split/empty
+  // rows — ops/csr_torch.build_worklist zero_rows).
   check_f32(x, "x");
+  check_f32(out, "out");
   TORCH_CHECK(wrow.scalar_type() == at::kInt &&
                   wbeg.scalar_type() == at::kLong &&
                   wave_start.scalar_type() == at::kInt &&
@@ -773,10 +776,6 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
               "worklist int32/int64 + indices int32 expected");
   const int F = x.size(1);
   const int n_waves = wave_start.numel() - 1;
-  at::Tensor out;
-  const bool acc = out_opt.has_value();
-  if (acc) { out = *out_opt; check_f32(out, "out"); }
-  else { out = at::zeros({n_rows, F}, x.options()); }
   if (wrow.numel() == 0 || n_waves <= 0) return out;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int grid = n_waves / 4;

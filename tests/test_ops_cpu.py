@@ -230,7 +230,9 @@ def test_build_worklist():
     from bnsgcn_amd.ops.csr_torch import build_worklist
     # degrees: 0, 3, 5000, 10 with seg=2048 -> row 2 split into 3 items
     indptr = torch.tensor([0, 0, 3, 5003, 5013], dtype=torch.int64)
-    wrow, wbeg, wend, wave_start = build_worklist(indptr, seg=2048)
+    wrow, wbeg, wend, wave_start, zero_rows = build_worklist(indptr, seg=2048)
+    # zero_rows = empty rows (row 0, deg 0) + split rows (row 2)
+    assert sorted(zero_rows.tolist()) == [0, 2]
     assert wrow.numel() == 1 + 3 + 1  # deg-0 row dropped
     # every edge covered exactly once, rows correct
     cover = torch.zeros(5013, dtype=torch.int32)
