@@ -280,12 +280,12 @@ def test_segment_sum_edges_and_bincount():
     row = torch.repeat_interleave(torch.arange(200), indptr[1:] - indptr[:-1])
     want = torch.zeros(200, H).index_add(0, row, grad)
     ip, ix = indptr.cuda(), indices.cuda()
-    got = ext.segment_sum_edges(*_worklist_of(ip), None, grad.cuda(), 200).cpu()
+    got = ext.segment_sum_edges(*_worklist_of(ip)[:4], None, grad.cuda(), 200).cpu()
     torch.testing.assert_close(got, want, rtol=1e-4, atol=1e-4)
     # g_el oracle via transpose + eperm
     tip, tix, eperm = transpose_csr(ip, ix, 150)
     want2 = torch.zeros(150, H).index_add(0, indices.long(), grad)
-    got2 = ext.segment_sum_edges(*_worklist_of(tip), eperm, grad.cuda(), 150).cpu()
+    got2 = ext.segment_sum_edges(*_worklist_of(tip)[:4], eperm, grad.cuda(), 150).cpu()
     torch.testing.assert_close(got2, want2, rtol=1e-4, atol=1e-4)
     # bincount kernel
     v = torch.randint(0, 150, (3000,), dtype=torch.int32).cuda()
