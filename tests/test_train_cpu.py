@@ -224,3 +224,34 @@ def test_fuzz_configs(tmp_path, seed):
         res = _run_config(tmp_path / f"f{seed}_{trial}", world, **kw)
         for m in res:
             assert np.isfinite(m["loss_history"]).all(), (kw, m)
+
+
+def test_main_cli_launcher(tmp_path):
+    """The reference-compatible `python main.py` entry: partitions and
+    spawns one process per partition (reference main.py:10-64)."""
+    import subprocess, sys
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    from util_dist import free_port
+    r = subprocess.run(
+        [sys.executable, os.path.join(os.path.dirname(__file__), "..", "main.py"),
+         "--dataset", "tiny", "--model", "graphsage", "--n-partitions", "2",
+         "--n-epochs", "2", "--n-hidden", "8", "--no-eval", "--backend", "gloo",
+         "--device", "cpu", "--use-pp", "--port", str(free_port()),
+         "--partition-dir", str(tmp_path / "p")],
+        capture_output=True, text=True, timeout=240, env=env,
+        cwd=str(tmp_path))
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_partition_cli(tmp_path):
+    import subprocess, sys
+    r = subprocess.run(
+        [sys.executable, os.path.join(os.path.dirname(__file__), "..",
+                                      "partition.py"),
+         "--dataset", "tiny", "--n-partitions", "3",
+         "--partition-dir", str(tmp_path / "p")],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (tmp_path / "p" / "tiny-3-metis-vol-trans" / "meta.json").exists()
