@@ -3,6 +3,7 @@ exchange must reproduce what the partition store already knows
 (reference protocol parity: get_boundary helper/utils.py:150-184,
 collect_out_degree train.py:148-167)."""
 import numpy as np
+import pytest
 import torch
 
 from bnsgcn_amd.graph import load_data, partition_graph
@@ -95,7 +96,7 @@ def test_syncbn_matches_single_process_batchnorm():
     torch.testing.assert_close(ours_unbiased, bn.running_var, rtol=1e-3, atol=1e-4)
 
 
-def _agg_rank(rank, world, parts_data, rate, n_epochs):
+def _agg_rank(rank, world, parts_data, rate, n_epochs, mode="mean"):
     """Average the sampled partition-aggregate over many epochs; the
     1/ratio-unbiased estimator must converge to the full (p=1.0)
     aggregation (reference estimator semantics,
@@ -110,24 +111,25 @@ def _agg_rank(rank, world, parts_data, rate, n_epochs):
     plan_full = HaloPlan(part, 1.0, seed=5, device="cpu")
     ctx_full = GraphContext.for_partition(part, plan_full, "cpu")
     plan_full.set_epoch(0)
-    exact = ctx_full.aggregate(feat, "mean")
+    exact = ctx_full.aggregate(feat, mode)
 
     plan = HaloPlan(part, rate, seed=5, device="cpu")
     ctx = GraphContext.for_partition(part, plan, "cpu")
     acc = torch.zeros_like(exact)
     for ep in range(n_epochs):
         plan.set_epoch(ep)
-        acc += ctx.aggregate(feat, "mean")
+        acc += ctx.aggregate(feat, mode)
     mean_est = acc / n_epochs
     err = (mean_est - exact).norm() / exact.norm()
     return float(err)
 
 
-def test_bns_estimator_unbiased():
+@pytest.mark.parametrize("mode", ["mean", "gcn"])
+def test_bns_estimator_unbiased(mode):
     from bnsgcn_amd.graph import load_data, partition_graph
     g = load_data("tiny", seed=21)
     parts, meta = partition_graph(g, 2, method="random", seed=1)
-    errs = run_dist(2, _agg_rank, (parts, 0.5, 300), timeout=600)
+    errs = run_dist(2, _agg_rank, (parts, 0.5, 300, mode), timeout=600)
     # Monte-Carlo average over 300 epochs: relative error shrinks well
     # below the single-sample deviation (~0.3 at p=0.5)
     assert max(errs) < 0.05, errs
