@@ -124,8 +124,7 @@ class GATLayer(nn.Module):
         if ctx.plan is None:                       # full-graph eval path
             ip, ix, tip, tix, eperm, _ = ctx.gat_block()
             z = F.linear(h, self.fc.weight).view(-1, H, D)
-            el = (z * self.attn_l).sum(-1)
-            er = (z * self.attn_r).sum(-1)
+            el, er = F.attn_project(z, self.attn_l, self.attn_r)
             logits = F.sddmm_add(el, er, ip, ix, tip, tix, eperm,
                                  slope=self.negative_slope)
             alpha = self.attn_drop(F.segment_softmax(logits, ip))
@@ -141,9 +140,8 @@ class GATLayer(nn.Module):
 
         z_in = F.linear(h, self.fc.weight).view(-1, H, D)
         z_h = F.linear(src_halo, self.fc.weight).view(-1, H, D)
-        el_in = (z_in * self.attn_l).sum(-1)       # inner sources
-        el_h = (z_h * self.attn_l).sum(-1)         # halo sources
-        er = (z_in * self.attn_r).sum(-1)          # inner destinations
+        el_in, er = F.attn_project(z_in, self.attn_l, self.attn_r)
+        el_h = (z_h * self.attn_l).sum(-1)         # halo sources (el only)
         slope = self.negative_slope
         li = F.sddmm_add(el_in, er, ctx.indptr, ctx.indices,
                          ctx.t_indptr, ctx.t_indices, ctx.t_eperm, slope=slope)

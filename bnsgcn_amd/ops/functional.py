@@ -303,3 +303,33 @@ class _SegmentSoftmax2(Function):
 
 def segment_softmax2(logits1, logits2, indptr1, indptr2):
     return _SegmentSoftmax2.apply(logits1, logits2, indptr1, indptr2)
+
+
+class _AttnProject(Function):
+    """Fused GAT attention projections (el, er) = (<z,a_l>, <z,a_r>) per
+    (node, head) — one read of z instead of torch's broadcast-mul +
+    reduce chains in each direction."""
+
+    @staticmethod
+    def forward(ctx, z, al, ar):
+        ctx.save_for_backward(z, al, ar)
+        if use_hip(z) and z.shape[1] * z.shape[2] % 4 == 0 and z.shape[1] <= 8:
+            el, er = get_ext().attn_project(z.contiguous(), al.contiguous(),
+                                            ar.contiguous())
+            return el, er
+        return ref.attn_project(z, al, ar)
+
+    @staticmethod
+    def backward(ctx, g_el, g_er):
+        z, al, ar = ctx.saved_tensors
+        g_el = g_el.contiguous()
+        g_er = g_er.contiguous()
+        if use_hip(z) and z.shape[1] * z.shape[2] % 4 == 0 and z.shape[1] <= 8:
+            dz, dal, dar = get_ext().attn_project_backward(
+                z.contiguous(), al.contiguous(), ar.contiguous(), g_el, g_er)
+            return dz, dal, dar
+        return ref.attn_project_backward(z, al, ar, g_el, g_er)
+
+
+def attn_project(z, al, ar):
+    return _AttnProject.apply(z, al, ar)

@@ -741,6 +741,93 @@ __global__ void bincount_i32_kernel(const int32_t* __restrict__ v, int64_t n,
     atomicAdd(reinterpret_cast<unsigned long long*>(&out[v[i]]), 1ull);
 }
 
+// Fused GAT attention projections: el[n,h] = <z[n,h,:], al[h,:]>,
+// er[n,h] = <z[n,h,:], ar[h,:]> — z is read ONCE (torch's broadcast-mul
+// + reduce chain reads/writes the [N,H,D] product twice per direction).
+// Wave per node row; per-lane partial accumulators indexed by the head a
+// float4 slot belongs to (H <= 8), wave-reduced per head.
+__global__ __launch_bounds__(256) void attn_project_fwd_kernel(
+    const float* __restrict__ z, const float* __restrict__ al,
+    const float* __restrict__ ar, float* __restrict__ el,
+    float* __restrict__ er, int64_t n, int H, int D, int hd4) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const float4* __restrict__ z4 = reinterpret_cast<const float4*>(z);
+  const float4* __restrict__ al4 = reinterpret_cast<const float4*>(al);
+  const float4* __restrict__ ar4 = reinterpret_cast<const float4*>(ar);
+  for (int64_t r = wave; r < n; r += n_waves) {
+    float accl[8], accr[8];
+#pragma unroll
+    for (int h = 0; h < 8; ++h) { accl[h] = 0.f; accr[h] = 0.f; }
+    for (int f = lane; f < hd4; f += WAVE) {
+      const int h = (f * 4) / D;
+      const float4 zv = z4[r * hd4 + f];
+      const float4 lv = al4[f];
+      const float4 rv = ar4[f];
+      accl[h] += zv.x * lv.x + zv.y * lv.y + zv.z * lv.z + zv.w * lv.w;
+      accr[h] += zv.x * rv.x + zv.y * rv.y + zv.z * rv.z + zv.w * rv.w;
+    }
+    for (int h = 0; h < H; ++h) {
+      const float sl = wave_reduce_sum(accl[h]);
+      const float sr = wave_reduce_sum(accr[h]);
+      if (lane == 0) {
+        el[r * H + h] = sl;
+        er[r * H + h] = sr;
+      }
+    }
+  }
+}
+
+// dz[n,h,d] = al[h,d]*g_el[n,h] + ar[h,d]*g_er[n,h] (single pass)
+__global__ __launch_bounds__(256) void attn_project_dz_kernel(
+    const float* __restrict__ al, const float* __restrict__ ar,
+    const float* __restrict__ g_el, const float* __restrict__ g_er,
+    float* __restrict__ dz, int64_t n, int H, int D, int hd4) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const float4* __restrict__ al4 = reinterpret_cast<const float4*>(al);
+  const float4* __restrict__ ar4 = reinterpret_cast<const float4*>(ar);
+  float4* __restrict__ dz4 = reinterpret_cast<float4*>(dz);
+  for (int64_t r = wave; r < n; r += n_waves) {
+    for (int f = lane; f < hd4; f += WAVE) {
+      const int h = (f * 4) / D;
+      const float gl = g_el[r * H + h];
+      const float gr = g_er[r * H + h];
+      const float4 lv = al4[f];
+      const float4 rv = ar4[f];
+      dz4[r * hd4 + f] = make_float4(lv.x * gl + rv.x * gr,
+                                     lv.y * gl + rv.y * gr,
+                                     lv.z * gl + rv.z * gr,
+                                     lv.w * gl + rv.w * gr);
+    }
+  }
+}
+
+// dal[h,d] = sum_n z[n,h,d]*g_el[n,h]; dar likewise — column reduction
+// with row chunks + atomics (out pre-zeroed by the launcher).
+__global__ void attn_project_dattn_kernel(
+    const float* __restrict__ z, const float* __restrict__ g_el,
+    const float* __restrict__ g_er, float* __restrict__ dal,
+    float* __restrict__ dar, int64_t n, int H, int D, int HD,
+    int row_chunks) {
+  const int f = blockIdx.x * blockDim.x + threadIdx.x;
+  if (f >= HD) return;
+  const int h = f / D;
+  const int64_t rows_per = (n + row_chunks - 1) / row_chunks;
+  const int64_t r0 = (int64_t)blockIdx.y * rows_per;
+  const int64_t r1 = (r0 + rows_per < n) ? r0 + rows_per : n;
+  float sl = 0.f, sr = 0.f;
+  for (int64_t r = r0; r < r1; ++r) {
+    const float zv = z[r * HD + f];
+    sl += zv * g_el[r * H + h];
+    sr += zv * g_er[r * H + h];
+  }
+  if (row_chunks == 1) { dal[f] = sl; dar[f] = sr; }
+  else { atomicAdd(&dal[f], sl); atomicAdd(&dar[f], sr); }
+}
+
 // ------------------------------ launchers ------------------------------
 
 inline void check_f32(const at::Tensor& t, const char* name) {
@@ -1049,6 +1136,51 @@ at::Tensor segment_sum_edges(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
   return out;
 }
 
+std::vector<at::Tensor> attn_project(at::Tensor z, at::Tensor al,
+                                     at::Tensor ar) {
+  check_f32(z, "z");
+  const int64_t n = z.size(0);
+  const int H = z.size(1), D = z.size(2);
+  TORCH_CHECK(H <= 8 && (H * D) % 4 == 0, "attn_project shape limits");
+  auto el = at::empty({n, H}, z.options());
+  auto er = at::empty({n, H}, z.options());
+  if (n == 0) return {el, er};
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(attn_project_fwd_kernel, dim3(spmm_grid((int)std::min<int64_t>(n, 1 << 28))),
+                     dim3(256), 0, stream, z.data_ptr<float>(),
+                     al.data_ptr<float>(), ar.data_ptr<float>(),
+                     el.data_ptr<float>(), er.data_ptr<float>(), n, H, D,
+                     H * D / 4);
+  return {el, er};
+}
+
+std::vector<at::Tensor> attn_project_backward(at::Tensor z, at::Tensor al,
+                                              at::Tensor ar, at::Tensor g_el,
+                                              at::Tensor g_er) {
+  check_f32(z, "z");
+  const int64_t n = z.size(0);
+  const int H = z.size(1), D = z.size(2);
+  const int HD = H * D;
+  auto dz = at::empty_like(z);
+  auto dal = at::zeros({1, H, D}, z.options());
+  auto dar = at::zeros({1, H, D}, z.options());
+  if (n == 0) return {dz, dal, dar};
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(attn_project_dz_kernel,
+                     dim3(spmm_grid((int)std::min<int64_t>(n, 1 << 28))),
+                     dim3(256), 0, stream, al.data_ptr<float>(),
+                     ar.data_ptr<float>(), g_el.data_ptr<float>(),
+                     g_er.data_ptr<float>(), dz.data_ptr<float>(), n, H, D,
+                     HD / 4);
+  const int row_chunks = (int)std::min<int64_t>((n + 255) / 256, 2048);
+  dim3 grid((HD + 255) / 256, row_chunks);
+  hipLaunchKernelGGL(attn_project_dattn_kernel, grid, dim3(256), 0, stream,
+                     z.data_ptr<float>(), g_el.data_ptr<float>(),
+                     g_er.data_ptr<float>(), dal.data_ptr<float>(),
+                     dar.data_ptr<float>(), n, H, D, HD, row_chunks);
+  return {dz, dal, dar};
+}
+
 at::Tensor gemm_strided(const at::Tensor& A, int64_t sAm, int64_t sAk,
                         const at::Tensor& B, int64_t sBk, int64_t sBn,
                         const c10::optional<at::Tensor>& bias,
@@ -1131,6 +1263,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("spmm_edge_sum", &spmm_edge_sum, "multi-head edge-weighted SpMM");
   m.def("sddmm_dot", &sddmm_dot, "per-edge per-head dot (spmm_edge grad)");
   m.def("bincount_i32", &bincount_i32, "atomic int32 histogram");
+  m.def("attn_project", &attn_project, "fused GAT el/er projections");
+  m.def("attn_project_backward", &attn_project_backward,
+        "fused GAT projection backward");
   m.def("segment_sum_edges", &segment_sum_edges,
         "segment sum of (permuted) edge values by CSR row");
 }

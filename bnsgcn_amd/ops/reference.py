@@ -170,3 +170,19 @@ def segment_softmax2_backward(indptr1, alpha1, grad1, indptr2, alpha2, grad2):
     d1 = alpha1 * (grad1 - s[row1]) if alpha1.numel() else alpha1
     d2 = alpha2 * (grad2 - s[row2]) if alpha2.numel() else alpha2
     return d1, d2
+
+
+def attn_project(z, al, ar):
+    """GAT attention projections in one conceptual pass:
+    el[n,h] = <z[n,h,:], al[h,:]>, er[n,h] = <z[n,h,:], ar[h,:]>.
+    z: [N,H,D]; al/ar: [1,H,D] (the GATConv attn_l/attn_r parameters)."""
+    el = (z * al).sum(-1)
+    er = (z * ar).sum(-1)
+    return el, er
+
+
+def attn_project_backward(z, al, ar, g_el, g_er):
+    dz = al * g_el.unsqueeze(-1) + ar * g_er.unsqueeze(-1)
+    dal = (z * g_el.unsqueeze(-1)).sum(0, keepdim=True)
+    dar = (z * g_er.unsqueeze(-1)).sum(0, keepdim=True)
+    return dz, dal, dar

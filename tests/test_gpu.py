@@ -377,3 +377,26 @@ def test_gemm_empty_inputs():
     assert ext.gemm_nn(g0, w).shape == (0, 300)
     dw = ext.gemm_tn(g0, x0)
     torch.testing.assert_close(dw, torch.zeros(64, 300).cuda())
+
+
+@needs_gpu
+@pytest.mark.parametrize("H,D", [(4, 128), (4, 100), (2, 8)])
+def test_attn_project_gpu(H, D):
+    from bnsgcn_amd.ops import functional as BF
+    z = torch.randn(500, H, D).cuda().requires_grad_(True)
+    al = torch.randn(1, H, D).cuda().requires_grad_(True)
+    ar = torch.randn(1, H, D).cuda().requires_grad_(True)
+    el, er = BF.attn_project(z, al, ar)
+    z2 = z.detach().clone().requires_grad_(True)
+    al2 = al.detach().clone().requires_grad_(True)
+    ar2 = ar.detach().clone().requires_grad_(True)
+    el2 = (z2 * al2).sum(-1)
+    er2 = (z2 * ar2).sum(-1)
+    torch.testing.assert_close(el, el2, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(er, er2, rtol=1e-4, atol=1e-4)
+    g1, g2 = torch.randn_like(el), torch.randn_like(er)
+    (el * g1 + er * g2).sum().backward()
+    (el2 * g1 + er2 * g2).sum().backward()
+    torch.testing.assert_close(z.grad, z2.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(al.grad, al2.grad, rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(ar.grad, ar2.grad, rtol=1e-3, atol=1e-3)

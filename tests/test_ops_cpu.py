@@ -334,3 +334,14 @@ def test_segment_softmax2_gradcheck():
     l1 = torch.randn(18, 2, dtype=torch.float64, requires_grad=True)
     l2 = torch.randn(9, 2, dtype=torch.float64, requires_grad=True)
     assert gradcheck(lambda a, b: F.segment_softmax2(a, b, ip1, ip2), (l1, l2))
+
+
+def test_attn_project_matches_torch():
+    from torch.autograd import gradcheck
+    z = torch.randn(20, 3, 8, dtype=torch.float64, requires_grad=True)
+    al = torch.randn(1, 3, 8, dtype=torch.float64, requires_grad=True)
+    ar = torch.randn(1, 3, 8, dtype=torch.float64, requires_grad=True)
+    el, er = F.attn_project(z, al, ar)
+    torch.testing.assert_close(el, (z * al).sum(-1))
+    torch.testing.assert_close(er, (z * ar).sum(-1))
+    assert gradcheck(lambda a, b, c: F.attn_project(a, b, c), (z, al, ar))
