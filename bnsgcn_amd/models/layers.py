@@ -141,7 +141,7 @@ class GATLayer(nn.Module):
         z_in = F.linear(h, self.fc.weight).view(-1, H, D)
         z_h = F.linear(src_halo, self.fc.weight).view(-1, H, D)
         el_in, er = F.attn_project(z_in, self.attn_l, self.attn_r)
-        el_h = (z_h * self.attn_l).sum(-1)         # halo sources (el only)
+        el_h, _ = F.attn_project(z_h, self.attn_l, self.attn_r)  # halo: el
         slope = self.negative_slope
         li = F.sddmm_add(el_in, er, ctx.indptr, ctx.indices,
                          ctx.t_indptr, ctx.t_indices, ctx.t_eperm, slope=slope)
