@@ -22,7 +22,8 @@ from .csr import Graph
 from .store import Partition, save_partitions
 
 
-def assign_parts(n_nodes: int, n_parts: int, method: str, seed: int = 0) -> np.ndarray:
+def assign_parts(n_nodes: int, n_parts: int, method: str, seed: int = 0,
+                 adj=None) -> np.ndarray:
     if method == "random":
         rng = np.random.default_rng(seed)
         part = np.arange(n_nodes, dtype=np.int32) % n_parts
@@ -34,14 +35,94 @@ def assign_parts(n_nodes: int, n_parts: int, method: str, seed: int = 0) -> np.n
         for p in range(n_parts):
             part[bounds[p]:bounds[p + 1]] = p
         return part
+    if method == "bfs":
+        assert adj is not None, "bfs partitioning needs the adjacency"
+        return _bfs_grow_parts(adj, n_parts, seed)
     raise ValueError(f"unknown partition method: {method}")
+
+
+def _bfs_grow_parts(adj, n_parts: int, seed: int) -> np.ndarray:
+    """Balanced multi-source BFS region growing — a cut-reducing
+    partitioner that does NOT rely on node-id locality (the contiguous
+    "metis" stand-in does; real-world node orderings may be arbitrary).
+    P seeds grow breadth-first with a node-count capacity of ~N/P each;
+    round-robin growth order rotates so no partition starves; leftover
+    (unreached/over-capacity) nodes are filled round-robin.
+
+    Vectorized per (partition, level): each expansion is one CSR gather
+    over the current frontier — O(E) total like a plain BFS.
+    """
+    rng = np.random.default_rng(seed)
+    n = adj.n_rows
+    indptr, indices = adj.indptr, adj.indices
+    cap = int(np.ceil(n / n_parts))
+    part = np.full(n, -1, dtype=np.int32)
+    sizes = np.zeros(n_parts, dtype=np.int64)
+    seeds = rng.choice(n, size=n_parts, replace=False)
+    frontiers: list[np.ndarray] = []
+    for p, s in enumerate(seeds):
+        part[s] = p
+        sizes[p] = 1
+        frontiers.append(np.array([s], dtype=np.int64))
+
+    active = True
+    order = list(range(n_parts))
+    while active:
+        active = False
+        for p in order:
+            if sizes[p] >= cap or len(frontiers[p]) == 0:
+                continue
+            f = frontiers[p]
+            # neighbors of the frontier (gather all adjacency rows)
+            lens = (indptr[f + 1] - indptr[f]).astype(np.int64)
+            total = int(lens.sum())
+            if total == 0:
+                frontiers[p] = np.zeros(0, dtype=np.int64)
+                continue
+            starts = indptr[f]
+            pos = np.arange(total)
+            row_of = np.repeat(np.arange(len(f)), lens)
+            offs = np.zeros(len(f) + 1, dtype=np.int64)
+            np.cumsum(lens, out=offs[1:])
+            nbr = indices[starts[row_of] + (pos - offs[row_of])].astype(np.int64)
+            nbr = np.unique(nbr)
+            nbr = nbr[part[nbr] < 0]
+            room = cap - int(sizes[p])
+            if len(nbr) > room:
+                nbr = nbr[:room]
+            if len(nbr):
+                part[nbr] = p
+                sizes[p] += len(nbr)
+                frontiers[p] = nbr
+                active = True
+            else:
+                frontiers[p] = np.zeros(0, dtype=np.int64)
+        order = order[1:] + order[:1]   # rotate growth priority
+
+    left = np.flatnonzero(part < 0)
+    if len(left):
+        # fill stragglers into the emptiest partitions
+        fill_order = np.argsort(sizes)
+        assign = np.empty(len(left), dtype=np.int32)
+        i = 0
+        for p in fill_order:
+            take = min(len(left) - i, max(0, cap - int(sizes[p])))
+            assign[i:i + take] = p
+            sizes[p] += take
+            i += take
+            if i >= len(left):
+                break
+        if i < len(left):
+            assign[i:] = np.arange(len(left) - i) % n_parts
+        part[left] = assign
+    return part
 
 
 def partition_graph(g: Graph, n_parts: int, method: str = "metis", seed: int = 0,
                     objective: str = "vol") -> tuple[list[Partition], dict]:
     """Split `g` into per-rank Partition objects (see store.Partition)."""
     n = g.n_nodes
-    part = assign_parts(n, n_parts, method, seed)
+    part = assign_parts(n, n_parts, method, seed, adj=g.adj_in)
 
     # inner-local id of every node within its partition (sorted-global order)
     inner_local = np.zeros(n, dtype=np.int64)

@@ -66,7 +66,7 @@ def test_multilabel_dataset():
     assert g.multilabel and g.label.shape == (g.n_nodes, 5)
 
 
-@pytest.mark.parametrize("method", ["random", "metis"])
+@pytest.mark.parametrize("method", ["random", "metis", "bfs"])
 def test_partition_invariants(method):
     g = load_data("tiny", seed=1)
     P = 4
@@ -126,3 +126,28 @@ def test_subgraph_inductive():
     eset = set(zip(gs.tolist(), gd.tolist()))
     for a, b in zip(keep[s].tolist(), keep[d].tolist()):
         assert (a, b) in eset
+
+
+def test_bfs_partitioner_cuts_less_than_random():
+    """BFS region growing must recover structure WITHOUT node-id locality:
+    a ring lattice under a random id permutation (contiguous/"metis"
+    partitioning is as bad as random there; BFS finds the arcs)."""
+    rng = np.random.default_rng(9)
+    n, k = 400, 4
+    base = np.arange(n)
+    src = np.concatenate([base] * (2 * k))
+    dst = np.concatenate([(base + off) % n
+                          for off in list(range(1, k + 1)) +
+                          list(range(-k, 0))])
+    perm = rng.permutation(n)
+    c = CSR.from_edges(perm[src], perm[dst], n, n)
+    from bnsgcn_amd.graph.partition import assign_parts
+    pb = assign_parts(n, 4, "bfs", seed=0, adj=c)
+    pr = assign_parts(n, 4, "random", seed=0)
+    s, d = c.to_edges()
+    cut_b = (pb[s] != pb[d]).mean()
+    cut_r = (pr[s] != pr[d]).mean()
+    assert cut_b < cut_r * 0.25, (cut_b, cut_r)
+    counts = np.bincount(pb, minlength=4)
+    assert counts.max() <= np.ceil(n / 4) + 1
+    assert counts.min() > 0
