@@ -255,3 +255,11 @@ def test_partition_cli(tmp_path):
         capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stderr[-2000:]
     assert (tmp_path / "p" / "tiny-3-metis-vol-trans" / "meta.json").exists()
+
+
+def test_norm_none_and_weight_decay(tmp_path):
+    multi = _run_config(tmp_path, 2, model="graphsage", sampling_rate=0.5,
+                        use_pp=True, n_epochs=6, norm="none",
+                        weight_decay=5e-4)
+    for m in multi:
+        assert np.isfinite(m["loss_history"]).all()
