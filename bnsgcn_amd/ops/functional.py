@@ -48,10 +48,16 @@ def spmm_edge_raw(indptr, indices, eweight, x, out=None, wperm=None):
     """wperm: optional edge permutation applied to eweight INSIDE the HIP
     kernel (fuses the w[eperm] gather of the transposed backward pass)."""
     if use_hip(x):
-        wl = _worklist_of(indptr)[:4]
-        return get_ext().spmm_edge_sum(*wl, indptr, indices,
-                                       eweight.contiguous(), wperm,
-                                       x.contiguous(), out)
+        wrow, wbeg, wend, wstart, zero_rows = _worklist_of(indptr)
+        acc = out is not None
+        if not acc:
+            out = torch.empty(indptr.numel() - 1, x.shape[1], x.shape[2],
+                              dtype=x.dtype, device=x.device)
+            if zero_rows.numel():
+                out.index_fill_(0, zero_rows, 0.0)
+        return get_ext().spmm_edge_sum(wrow, wbeg, wend, wstart, indptr,
+                                       indices, eweight.contiguous(), wperm,
+                                       x.contiguous(), out, acc)
     ew = eweight if wperm is None else eweight[wperm]
     return ref.spmm_edge_sum(indptr, indices, ew, x, out)
 

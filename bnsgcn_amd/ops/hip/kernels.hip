@@ -1054,14 +1054,11 @@ at::Tensor spmm_edge_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                          at::Tensor wstart, at::Tensor indptr,
                          at::Tensor indices, at::Tensor w,
                          c10::optional<at::Tensor> wperm, at::Tensor x,
-                         c10::optional<at::Tensor> out_opt) {
-  check_f32(w, "w"); check_f32(x, "x");
+                         at::Tensor out, bool acc) {
+  // out is caller-allocated (split/empty rows pre-zeroed — see spmm_sum)
+  check_f32(w, "w"); check_f32(x, "x"); check_f32(out, "out");
   const int n_rows = indptr.numel() - 1;
   const int H = x.size(1), D = x.size(2);
-  at::Tensor out;
-  const bool acc = out_opt.has_value();
-  if (acc) out = *out_opt;
-  else out = at::zeros({n_rows, H, D}, x.options());
   if (indices.numel() == 0 || n_rows == 0) return out;
   auto stream = at::cuda::getCurrentCUDAStream();
   const int n_waves = wstart.numel() - 1;
