@@ -187,18 +187,49 @@ __global__ void pack_rows_kernel(const float* __restrict__ x,
   }
 }
 
+__global__ void pack_rows_vec4_kernel(const float* __restrict__ x,
+                                      const int64_t* __restrict__ idx,
+                                      const float* __restrict__ scale,
+                                      float* __restrict__ out, int n, int f4) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t total = (int64_t)n * f4;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float4* __restrict__ x4 = reinterpret_cast<const float4*>(x);
+  float4* __restrict__ out4 = reinterpret_cast<float4*>(out);
+  for (int64_t i = t; i < total; i += stride) {
+    const int r = i / f4, f = i - (int64_t)r * f4;
+    const float s = scale ? scale[r] : 1.0f;
+    const float4 v = x4[idx[r] * f4 + f];
+    out4[i] = make_float4(s * v.x, s * v.y, s * v.z, s * v.w);
+  }
+}
+
 __global__ void scatter_add_rows_kernel(float* __restrict__ out,
                                         const int64_t* __restrict__ idx,
                                         const float* __restrict__ src,
                                         const float* __restrict__ scale,
                                         int n, int F) {
+  // 4 floats per thread (vectorized read; atomics stay per-float —
+  // destination rows can repeat across peers)
   const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  const int64_t total = (int64_t)n * F;
+  const int64_t q = F / 4;
+  const bool vec = (F % 4 == 0);
+  const int64_t total = vec ? (int64_t)n * q : (int64_t)n * F;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float4* __restrict__ src4 = reinterpret_cast<const float4*>(src);
   for (int64_t i = t; i < total; i += stride) {
-    const int r = i / F, f = i - (int64_t)r * F;
-    const float s = scale ? scale[r] : 1.0f;
-    atomicAdd(&out[idx[r] * F + f], s * src[i]);
+    if (vec) {
+      const int r = i / q, f = i - (int64_t)r * q;
+      const float s = scale ? scale[r] : 1.0f;
+      const float4 v = src4[i];
+      float* p = &out[idx[r] * F + f * 4];
+      atomicAdd(p + 0, s * v.x); atomicAdd(p + 1, s * v.y);
+      atomicAdd(p + 2, s * v.z); atomicAdd(p + 3, s * v.w);
+    } else {
+      const int r = i / F, f = i - (int64_t)r * F;
+      const float s = scale ? scale[r] : 1.0f;
+      atomicAdd(&out[idx[r] * F + f], s * src[i]);
+    }
   }
 }
 
@@ -908,9 +939,15 @@ at::Tensor pack_rows(at::Tensor x, at::Tensor idx,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int64_t total = (int64_t)n * F;
   const int grid = std::min<int64_t>((total + 255) / 256, 4096);
-  hipLaunchKernelGGL(pack_rows_kernel, dim3(grid), dim3(256), 0, stream,
-                     x.data_ptr<float>(), idx.data_ptr<int64_t>(),
-                     opt_ptr(scale), out.data_ptr<float>(), n, F);
+  if (F % 4 == 0) {
+    hipLaunchKernelGGL(pack_rows_vec4_kernel, dim3(grid), dim3(256), 0, stream,
+                       x.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                       opt_ptr(scale), out.data_ptr<float>(), n, F / 4);
+  } else {
+    hipLaunchKernelGGL(pack_rows_kernel, dim3(grid), dim3(256), 0, stream,
+                       x.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                       opt_ptr(scale), out.data_ptr<float>(), n, F);
+  }
   return out;
 }
 
