@@ -74,18 +74,19 @@ def test_spmm_heavy_row_split():
 
 
 @needs_gpu
-def test_pack_scatter_match():
+@pytest.mark.parametrize("Fdim", [64, 31])
+def test_pack_scatter_match(Fdim):
     torch.manual_seed(1)
-    x = torch.randn(300, 64)
+    x = torch.randn(300, Fdim)
     idx = torch.randint(0, 300, (120,))
     scale = torch.rand(120) + 0.5
     want = ref.pack_rows(x, idx, scale)
     got = ext.pack_rows(x.cuda(), idx.cuda(), scale.cuda()).cpu()
     torch.testing.assert_close(got, want)
-    out_c = torch.zeros(300, 64)
-    src = torch.randn(120, 64)
+    out_c = torch.zeros(300, Fdim)
+    src = torch.randn(120, Fdim)
     ref.scatter_add_rows(out_c, idx, src, scale)
-    out_g = torch.zeros(300, 64).cuda()
+    out_g = torch.zeros(300, Fdim).cuda()
     ext.scatter_add_rows(out_g, idx.cuda(), src.cuda(), scale.cuda())
     torch.testing.assert_close(out_g.cpu(), out_c, rtol=1e-5, atol=1e-5)
 
