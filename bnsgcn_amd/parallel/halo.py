@@ -140,7 +140,8 @@ class _HaloExchange(Function):
     @staticmethod
     def forward(ctx, x, plan: HaloPlan, st: EpochState):
         send = pack_rows_raw(x, st.pack_idx, st.pack_scale)
-        recv = _exchange(send, st.recv_counts, st.send_counts)
+        with comm_timer.span("forward", cuda=x.is_cuda):
+            recv = _exchange(send, st.recv_counts, st.send_counts)
         ctx.st = st
         ctx.n_inner = x.shape[0]
         return recv
@@ -149,7 +150,8 @@ class _HaloExchange(Function):
     def backward(ctx, g):
         st: EpochState = ctx.st
         g = g.contiguous()
-        back = _exchange(g, st.send_counts, st.recv_counts)
+        with comm_timer.span("backward", cuda=g.is_cuda):
+            back = _exchange(g, st.send_counts, st.recv_counts)
         gx = torch.zeros(ctx.n_inner, g.shape[1], dtype=g.dtype, device=g.device)
         scatter_add_rows_raw(gx, st.pack_idx, back, st.pack_scale)
         return gx, None, None

@@ -225,6 +225,14 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
     else:
         device = args.device
     if str(device).startswith("cuda"):
+        import torch.distributed as _d
+        if (_d.is_initialized() and _d.get_backend() == "nccl"
+                and world > torch.cuda.device_count()):
+            raise RuntimeError(
+                f"{world} ranks > {torch.cuda.device_count()} GPUs: RCCL "
+                "cannot oversubscribe devices (the reference's gloo backend "
+                "could, main.py:45). Use --n-partitions <= GPU count, or "
+                "--backend gloo --device cpu.")
         torch.cuda.set_device(torch.device(device))
 
     name = graph_name_of(args)

@@ -263,3 +263,16 @@ def test_norm_none_and_weight_decay(tmp_path):
                         weight_decay=5e-4)
     for m in multi:
         assert np.isfinite(m["loss_history"]).all()
+
+
+def test_gat_with_eval(tmp_path):
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        multi = _run_config(tmp_path, 2, model="gat", heads=2, n_hidden=16,
+                            sampling_rate=1.0, n_epochs=60, log_every=30,
+                            eval=True, lr=0.03)
+        assert "test_acc" in multi[0]
+        assert multi[0]["test_acc"] > 0.2   # above the 1/7 chance level
+    finally:
+        os.chdir(cwd)
