@@ -16,7 +16,7 @@ def create_parser() -> argparse.ArgumentParser:
                    choices=["gcn", "graphsage", "gat"])
     p.add_argument("--dropout", type=float, default=0.5)
     p.add_argument("--lr", type=float, default=1e-2)
-    p.add_argument("--sampling-rate", "--sampling_rate", type=float, default=0.1)
+    p.add_argument("--sampling-rate", "--sampling_rate", type=float, default=1.0)
     p.add_argument("--heads", type=int, default=1)
     p.add_argument("--n-epochs", "--n_epochs", type=int, default=200)
     p.add_argument("--n-partitions", "--n_partitions", type=int, default=2)
@@ -44,8 +44,12 @@ def create_parser() -> argparse.ArgumentParser:
     p.add_argument("--eval", action="store_true", dest="eval",
                    default=True, help="enable evaluation (default)")
     p.add_argument("--no-eval", action="store_false", dest="eval")
-    p.add_argument("--partition-dir", "--partition_dir", type=str,
+    p.add_argument("--partition-dir", "--partition_dir", "--part-path",
+                   "--part_path", dest="partition_dir", type=str,
                    default="partition")
+    p.add_argument("--data-path", "--data_path", type=str, default="./dataset/",
+                   help="dataset storage dir (reference parity; holds the "
+                        "synthetic edge cache here)")
     # --- bnsgcn_amd extras ---
     p.add_argument("--data-scale", type=float, default=1.0,
                    help="shrink the synthetic dataset (papers100M smoke runs)")

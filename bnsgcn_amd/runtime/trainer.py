@@ -42,7 +42,8 @@ def _load_data_cached(args):
     scaling sweep) skip the expensive edge sampling."""
     import numpy as np
     from ..graph.csr import CSR, Graph
-    cache = os.path.join(args.partition_dir,
+    cache_dir = getattr(args, "data_path", None) or args.partition_dir
+    cache = os.path.join(cache_dir,
                          f"_edges_{args.dataset}_s{args.seed}"
                          f"_x{args.data_scale:g}.npz")
     if os.path.exists(cache):
@@ -54,7 +55,7 @@ def _load_data_cached(args):
     else:
         g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
         try:
-            os.makedirs(args.partition_dir, exist_ok=True)
+            os.makedirs(cache_dir, exist_ok=True)
             np.savez(cache, indptr=g.adj_in.indptr, indices=g.adj_in.indices,
                      n_nodes=g.n_nodes, feat=g.feat, label=g.label,
                      train_mask=g.train_mask, val_mask=g.val_mask,
