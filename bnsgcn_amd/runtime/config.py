@@ -63,6 +63,12 @@ def create_parser() -> argparse.ArgumentParser:
                    help="checkpoint (.pth.tar state_dict) to load before "
                         "training — the reference saves checkpoints but has "
                         "no resume path (SURVEY.md §5.4); this adds one")
+    p.add_argument("--eval-mode", choices=["thread", "dist"], default="thread",
+                   help="'thread' = reference-style rank-0 full-graph eval in "
+                        "a background thread; 'dist' = exact p=1.0 eval-mode "
+                        "forward ACROSS the training partitions with "
+                        "all-reduced accuracy (transductive only, runs on "
+                        "the GPUs — no CPU full-graph pass)")
     p.add_argument("--eval-device", type=str, default="cpu",
                    help="device for rank-0 full-graph evaluation (the "
                         "reference evaluates on CPU; 'cuda' runs it on the "

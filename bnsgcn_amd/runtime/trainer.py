@@ -98,6 +98,9 @@ class RankState:
         lab = torch.from_numpy(part.label)
         self.label = lab.to(dev)
         self.train_mask = torch.from_numpy(part.train_mask).to(dev)
+        self.val_mask = torch.from_numpy(part.val_mask).to(dev)
+        self.test_mask = torch.from_numpy(part.test_mask).to(dev)
+        self.raw_feat = self.feat   # pre-precompute features (dist eval)
         self.plan = HaloPlan(part, args.sampling_rate, seed=args.seed, device=dev,
                              unit_ratio=(args.model == "gat"
                                          and not args.gat_ratio_scale))
@@ -118,6 +121,7 @@ class RankState:
         recv = torch.empty(sum(st.recv_counts), self.feat.shape[1],
                            dtype=self.feat.dtype, device=self.feat.device)
         all_to_all_rows(recv, send, st.recv_counts, st.send_counts)
+        self.raw_feat = self.feat
         if args.model == "gcn":
             out = spmm_sum_raw(ctx.indptr, ctx.indices, self.feat,
                                ctx.out_norm_inv, ctx.in_norm_inv)
@@ -135,6 +139,56 @@ class RankState:
             self.halo_feat0 = recv   # keep raw full-halo features
         else:
             raise ValueError(args.model)
+
+
+@torch.no_grad()
+def dist_evaluate(state: RankState, model) -> dict:
+    """Exact full-graph evaluation ACROSS the training partitions
+    (transductive only): every rank runs an eval-mode forward with the
+    FULL (p=1.0) halo state on its own partition, then correctness counts
+    are all-reduced. Replaces the reference's rank-0 CPU full-graph pass
+    (train.py:434-442) with a collective that runs on the GPUs in
+    milliseconds. Must be called by ALL ranks."""
+    assert not state.args.inductive, "dist eval is transductive-only"
+    plan = state.plan
+    saved = plan._state
+    plan._state = state.ctx.full_state()
+    was_training = model.training
+    try:
+        model.eval()
+        logits = model(state.ctx, state.raw_feat)
+    finally:
+        plan._state = saved
+        if was_training:
+            model.train()
+    multilabel = bool(state.part.meta.get("multilabel", False))
+    out = {}
+    if multilabel:
+        pred = (logits > 0).float()
+        lab = state.label.float()
+        stats = []
+        for m in (state.val_mask, state.test_mask):
+            tp = (pred[m] * lab[m]).sum()
+            fp = (pred[m] * (1 - lab[m])).sum()
+            fn = ((1 - pred[m]) * lab[m]).sum()
+            stats.append(torch.stack([tp, fp, fn]))
+        t = torch.cat(stats)
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.all_reduce(t)
+        for key, (tp, fp, fn) in zip(("val", "test"), t.view(2, 3)):
+            out[key] = float(2 * tp / (2 * tp + fp + fn + 1e-12))
+    else:
+        pred = logits.argmax(1)
+        t = torch.stack([
+            (pred[state.val_mask] == state.label[state.val_mask]).sum(),
+            state.val_mask.sum(),
+            (pred[state.test_mask] == state.label[state.test_mask]).sum(),
+            state.test_mask.sum()]).float()
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.all_reduce(t)
+        out["val"] = float(t[0] / t[1].clamp_min(1))
+        out["test"] = float(t[2] / t[3].clamp_min(1))
+    return out
 
 
 def _forward(model, state: RankState, feat):
@@ -278,9 +332,18 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
     pool = None
     pending = None
     best_val, best_state = -1.0, None
+    best_dist_test = None
+    eval_mode = getattr(args, "eval_mode", "thread")
+    if eval_mode == "dist" and args.inductive:
+        if rank == 0:
+            print("dist eval is transductive-only; falling back to thread "
+                  "mode", flush=True)
+        eval_mode = "thread"
+    dist_eval_on = args.eval and eval_mode == "dist"
     if args.eval and rank == 0:
-        evaluator = Evaluator(args)
-        pool = ThreadPoolExecutor(max_workers=1)
+        if not dist_eval_on:
+            evaluator = Evaluator(args)
+            pool = ThreadPoolExecutor(max_workers=1)
         os.makedirs("checkpoint", exist_ok=True)
         os.makedirs("results", exist_ok=True)
 
@@ -320,6 +383,19 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
                   f"Reduce(s) {np.mean(reduce_dur):.4f} | "
                   f"Loss {loss_val / max(state.part_train, 1):.4f}", flush=True)
 
+        if dist_eval_on and (epoch + 1) % args.log_every == 0:
+            # COLLECTIVE: every rank participates
+            res = dist_evaluate(state, model)
+            if rank == 0:
+                torch.save(model.state_dict(),
+                           f"checkpoint/{name}_p{args.sampling_rate:.2f}"
+                           f"_{epoch}.pth.tar")
+                if res["val"] > best_val:
+                    best_val = res["val"]
+                    best_dist_test = res["test"]
+                    best_state = {k: v.cpu() for k, v in
+                                  model.state_dict().items()}
+
         if evaluator is not None and (epoch + 1) % args.log_every == 0:
             torch.save(model.state_dict(),
                        f"checkpoint/{name}_p{args.sampling_rate:.2f}_{epoch}.pth.tar")
@@ -344,6 +420,15 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
                "comm_time": float(np.mean(comm_dur)) if comm_dur else 0.0,
                "loss": loss_val / max(state.part_train, 1),
                "loss_history": loss_history}
+    if dist_eval_on and rank == 0 and best_state is not None:
+        torch.save(best_state, f"checkpoint/{name}_final.pth.tar")
+        summary["val_acc"] = best_val
+        summary["test_acc"] = best_dist_test
+        with open(f"results/{args.dataset}_n{args.n_partitions}"
+                  f"_p{args.sampling_rate:.2f}.txt", "a") as f:
+            f.write(f"val={best_val:.4f} test={best_dist_test:.4f}\n")
+        print(f"Max Validation Accuracy {best_val:.4f} | "
+              f"Test Accuracy {best_dist_test:.4f}", flush=True)
     if evaluator is not None:
         if pending is not None:
             res, snap = pending.result()

@@ -281,3 +281,78 @@ def test_gat_with_eval(tmp_path):
         assert multi[0]["loss"] < 0.5
     finally:
         os.chdir(cwd)
+
+
+def _dist_eval_check(rank, world, args):
+    """Train briefly, then compare dist_evaluate against the rank-0
+    full-graph Evaluator on the SAME weights — must agree exactly."""
+    import torch
+    from bnsgcn_amd.runtime.trainer import (run, dist_evaluate, Evaluator,
+                                            RankState, _forward)
+    from bnsgcn_amd.graph import load_partition
+    from bnsgcn_amd.models.models import create_model
+    from bnsgcn_amd.parallel import init_distributed, GradReducer
+    os.chdir(os.path.dirname(args.partition_dir))
+    init_distributed("gloo", rank, world)
+    part = load_partition(args.partition_dir, args.graph_name, rank)
+    torch.manual_seed(args.seed)
+    state = RankState(part, args, "cpu")
+    state.plan.set_epoch(0)
+    model = create_model(args, n_feat=int(part.meta["n_feat"]),
+                         n_class=int(part.meta["n_class"]),
+                         train_size=int(part.meta["n_train"]))
+    if args.use_pp:
+        state.precompute()
+    import torch.distributed as dist
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    red = GradReducer(model, int(part.meta["n_train"]))
+    opt = torch.optim.Adam(model.parameters(), lr=0.05)
+    lf = torch.nn.CrossEntropyLoss(reduction="sum")
+    for ep in range(30):
+        state.plan.set_epoch(ep)
+        model.train()
+        loss = lf(_forward(model, state, state.feat)[state.train_mask],
+                  state.label[state.train_mask].long())
+        red.zero_grad()
+        loss.backward()
+        red.reduce()
+        red.synchronize()
+        opt.step()
+    res = dist_evaluate(state, model)
+    if rank == 0:
+        ev = Evaluator(args)
+        ref_model = create_model(args, n_feat=int(part.meta["n_feat"]),
+                                 n_class=int(part.meta["n_class"]),
+                                 train_size=int(part.meta["n_train"]))
+        ref_model.load_state_dict(model.state_dict())
+        want = ev.evaluate(ref_model)
+        return {"dist": res, "ref": want}
+    return {"dist": res}
+
+
+@pytest.mark.parametrize("model,use_pp", [("graphsage", True), ("gcn", False),
+                                          ("gat", False)])
+def test_dist_eval_matches_rank0_evaluator(tmp_path, model, use_pp):
+    args = make_args(tmp_path, n_partitions=2, model=model, use_pp=use_pp,
+                     sampling_rate=0.5, n_hidden=16, heads=2)
+    prepare_partitions(args)
+    res = run_dist(2, _dist_eval_check, (args,))
+    got, want = res[0]["dist"], res[0]["ref"]
+    assert abs(got["val"] - want["val"]) < 1e-5, (got, want)
+    assert abs(got["test"] - want["test"]) < 1e-5, (got, want)
+    # both ranks agree
+    assert abs(res[0]["dist"]["val"] - res[1]["dist"]["val"]) < 1e-9
+
+
+def test_run_with_dist_eval(tmp_path):
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        multi = _run_config(tmp_path, 2, model="graphsage", use_pp=True,
+                            sampling_rate=0.5, n_epochs=100, log_every=25,
+                            eval=True, eval_mode="dist", lr=0.05, n_hidden=32)
+        assert multi[0]["test_acc"] > 0.30
+        assert os.path.exists("checkpoint/tiny-2-metis-vol-trans_final.pth.tar")
+    finally:
+        os.chdir(cwd)
