@@ -335,6 +335,10 @@ def test_train_and_eval_on_gpu_accuracy():
         loss.backward()
         reducer.synchronize()
         opt.step()
+    # dist_evaluate on GPU (world=1): exercises the full-state eval forward
+    from bnsgcn_amd.runtime.trainer import dist_evaluate
+    dres = dist_evaluate(state, m)
+    assert 0.0 <= dres["test"] <= 1.0
     args.seed = 7
     ev = Evaluator(args)
     final = create_model(args, n_feat=g.n_feat, n_class=g.n_class,
@@ -342,6 +346,8 @@ def test_train_and_eval_on_gpu_accuracy():
     final.load_state_dict({k: v.cpu() for k, v in m.state_dict().items()})
     out = ev.evaluate(final)
     assert out["test"] > 0.30, out
+    # GPU dist eval == CPU full-graph evaluator on the same weights
+    assert abs(dres["test"] - out["test"]) < 1e-4, (dres, out)
 
 
 @needs_gpu
