@@ -29,20 +29,33 @@ class _SyncBNFunc(Function):
         else:
             from ..ops._ext import use_hip, get_ext
             if use_hip(x):
-                stats = get_ext().syncbn_stats(x.contiguous())
+                sums = get_ext().syncbn_stats(x.contiguous())
             else:
-                stats = torch.stack((x.sum(0), (x * x).sum(0)))
+                sums = torch.stack((x.sum(0), (x * x).sum(0)))
+            # The global row count rides along in the same all-reduce. The
+            # reference divides by the CONSTRUCTION-TIME whole_size
+            # (= n_train), which is the true row count only in inductive
+            # mode — transductively its variance goes negative -> NaN
+            # (SURVEY.md §2.5.8). Counting rows dynamically is identical
+            # inductively and correct transductively.
+            stats = torch.cat([sums.flatten(),
+                               sums.new_tensor([float(x.shape[0])])])
             _maybe_all_reduce(stats)
-            mean = stats[0] / whole_size
-            var = (stats[1] - mean * stats[0]) / whole_size
+            nf = x.shape[1]
+            n_total = stats[-1].clamp_min(1.0)
+            sum_x = stats[:nf]
+            sum_x2 = stats[nf:2 * nf]
+            mean = sum_x / n_total
+            var = (sum_x2 - mean * sum_x) / n_total
             with torch.no_grad():
                 running_mean.mul_(1 - momentum).add_(mean * momentum)
                 running_var.mul_(1 - momentum).add_(var * momentum)
-        std = torch.sqrt(var + eps)
+        std = torch.sqrt(var.clamp_min(0.0) + eps) if training \
+            else torch.sqrt(var + eps)
         x_hat = (x - mean) / std
         if training:
             ctx.save_for_backward(x_hat, weight, std)
-            ctx.whole_size = whole_size
+            ctx.whole_size = float(n_total)
         return x_hat * weight + bias
 
     @staticmethod

@@ -356,3 +356,15 @@ def test_run_with_dist_eval(tmp_path):
         assert os.path.exists("checkpoint/tiny-2-metis-vol-trans_final.pth.tar")
     finally:
         os.chdir(cwd)
+
+
+def test_syncbn_transductive_finite(tmp_path):
+    """Transductive + --norm batch: the reference divides BN stats by
+    n_train, whose variance goes NEGATIVE when partitions hold more rows
+    than train nodes (NaN loss — SURVEY.md §2.5.8). Our dynamic row count
+    must keep this finite."""
+    multi = _run_config(tmp_path, 2, model="gcn", sampling_rate=0.33,
+                        norm="batch", inductive=False, use_pp=False,
+                        n_epochs=6, n_hidden=33, n_layers=3)
+    for m in multi:
+        assert np.isfinite(m["loss_history"]).all()
