@@ -407,3 +407,25 @@ def test_attn_project_gpu(H, D):
     torch.testing.assert_close(z.grad, z2.grad, rtol=1e-4, atol=1e-4)
     torch.testing.assert_close(al.grad, al2.grad, rtol=1e-3, atol=1e-3)
     torch.testing.assert_close(ar.grad, ar2.grad, rtol=1e-3, atol=1e-3)
+
+
+@needs_gpu
+def test_syncbn_module_gpu_matches_cpu():
+    from bnsgcn_amd.models.sync_bn import SyncBatchNorm
+    torch.manual_seed(3)
+    x = torch.randn(500, 24)
+    def run(dev):
+        torch.manual_seed(4)
+        bn = SyncBatchNorm(24, whole_size=500).to(dev)
+        bn.train()
+        xx = x.clone().to(dev).requires_grad_(True)
+        y = bn(xx)
+        y.sum().backward()
+        return (y.detach().cpu(), xx.grad.cpu(), bn.weight.grad.cpu(),
+                bn.running_var.cpu())
+    yc, gc, wc, rvc = run("cpu")
+    yg, gg, wg, rvg = run("cuda:0")
+    torch.testing.assert_close(yg, yc, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(gg, gc, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(wg, wc, rtol=1e-4, atol=1e-3)
+    torch.testing.assert_close(rvg, rvc, rtol=1e-4, atol=1e-4)
