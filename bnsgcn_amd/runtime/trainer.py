@@ -67,20 +67,34 @@ def _load_data_cached(args):
 
 
 def prepare_partitions(args) -> str:
-    """Rank-0 offline step (reference main.py:26-31 + graph_partition)."""
+    """Rank-0 offline step (reference main.py:26-31 + graph_partition).
+
+    With --eval-mode dist in the INDUCTIVE setting, two extra partition
+    stores are written ("<name>-evalval" = the train∪val graph,
+    "<name>-evaltest" = the full graph) so evaluation can also run as a
+    distributed p=1.0 forward instead of the rank-0 CPU pass."""
     name = graph_name_of(args)
     d = os.path.join(args.partition_dir, name)
     if args.skip_partition and os.path.exists(os.path.join(d, "meta.json")):
         return d
-    g = _load_data_cached(args)
-    if args.inductive:
-        g = g.subgraph(g.train_mask, name=g.name)
+    g_full = _load_data_cached(args)
+    g = g_full.subgraph(g_full.train_mask, name=g_full.name) \
+        if args.inductive else g_full
     extra = {"inductive": args.inductive, "dataset_seed": args.seed,
              "data_scale": args.data_scale,
              "full_n_nodes": g.n_nodes, "full_n_edges": g.n_edges}
     partition_and_save(g, args.n_partitions, args.partition_method,
                        args.partition_dir, name, seed=args.seed,
                        objective=args.partition_obj, extra_meta=extra)
+    if (args.inductive and args.eval
+            and getattr(args, "eval_mode", "thread") == "dist"):
+        g_val = g_full.subgraph(g_full.train_mask | g_full.val_mask,
+                                name=g_full.name)
+        for suffix, gg in (("-evalval", g_val), ("-evaltest", g_full)):
+            partition_and_save(gg, args.n_partitions, args.partition_method,
+                               args.partition_dir, name + suffix,
+                               seed=args.seed, objective=args.partition_obj,
+                               extra_meta=extra)
     return d
 
 
@@ -141,15 +155,36 @@ class RankState:
             raise ValueError(args.model)
 
 
+class EvalState:
+    """Lightweight per-rank state over an EVALUATION partition store (the
+    train∪val graph / the full graph in the inductive setting): features,
+    labels, masks, context + a p=1.0 plan. Used by dist_evaluate."""
+
+    def __init__(self, part: Partition, args, device):
+        self.device = torch.device(device)
+        self.part = part
+        self.args = args
+        self.feat = torch.from_numpy(part.feat).to(device)
+        self.raw_feat = self.feat
+        self.label = torch.from_numpy(part.label).to(device)
+        self.val_mask = torch.from_numpy(part.val_mask).to(device)
+        self.test_mask = torch.from_numpy(part.test_mask).to(device)
+        self.plan = HaloPlan(part, 1.0, seed=args.seed, device=device,
+                             unit_ratio=True)
+        self.plan.set_epoch(0)
+        self.ctx = GraphContext.for_partition(part, self.plan, device)
+
+
 @torch.no_grad()
-def dist_evaluate(state: RankState, model) -> dict:
-    """Exact full-graph evaluation ACROSS the training partitions
-    (transductive only): every rank runs an eval-mode forward with the
-    FULL (p=1.0) halo state on its own partition, then correctness counts
-    are all-reduced. Replaces the reference's rank-0 CPU full-graph pass
-    (train.py:434-442) with a collective that runs on the GPUs in
-    milliseconds. Must be called by ALL ranks."""
-    assert not state.args.inductive, "dist eval is transductive-only"
+def dist_evaluate(state, model) -> dict:
+    """Exact full-graph evaluation ACROSS partitions: every rank runs an
+    eval-mode forward with the FULL (p=1.0) halo state on its own
+    partition, then correctness counts are all-reduced. Replaces the
+    reference's rank-0 CPU full-graph pass (train.py:434-442) with a
+    collective that runs on the GPUs in milliseconds. Must be called by
+    ALL ranks. `state` is the training RankState (transductive: the
+    training partitions ARE the full graph) or an EvalState over a
+    dedicated eval store (inductive)."""
     plan = state.plan
     saved = plan._state
     plan._state = state.ctx.full_state()
@@ -334,12 +369,23 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
     best_val, best_state = -1.0, None
     best_dist_test = None
     eval_mode = getattr(args, "eval_mode", "thread")
-    if eval_mode == "dist" and args.inductive:
-        if rank == 0:
-            print("dist eval is transductive-only; falling back to thread "
-                  "mode", flush=True)
-        eval_mode = "thread"
     dist_eval_on = args.eval and eval_mode == "dist"
+    eval_states = None
+    if dist_eval_on and args.inductive:
+        # inductive: evaluation graphs differ from the training graph —
+        # load the dedicated eval stores written by prepare_partitions
+        try:
+            pv = load_partition(args.partition_dir, name + "-evalval", rank)
+            pt = load_partition(args.partition_dir, name + "-evaltest", rank)
+            eval_states = (EvalState(pv, args, device),
+                           EvalState(pt, args, device))
+        except FileNotFoundError:
+            if rank == 0:
+                print("inductive dist eval stores missing (re-run "
+                      "partitioning with --eval-mode dist); falling back "
+                      "to thread mode", flush=True)
+            eval_mode = "thread"
+            dist_eval_on = False
     if args.eval and rank == 0:
         if not dist_eval_on:
             evaluator = Evaluator(args)
@@ -385,7 +431,11 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
 
         if dist_eval_on and (epoch + 1) % args.log_every == 0:
             # COLLECTIVE: every rank participates
-            res = dist_evaluate(state, model)
+            if eval_states is None:          # transductive: training
+                res = dist_evaluate(state, model)    # partitions = full graph
+            else:                            # inductive: dedicated stores
+                res = {"val": dist_evaluate(eval_states[0], model)["val"],
+                       "test": dist_evaluate(eval_states[1], model)["test"]}
             if rank == 0:
                 torch.save(model.state_dict(),
                            f"checkpoint/{name}_p{args.sampling_rate:.2f}"

@@ -368,3 +368,36 @@ def test_syncbn_transductive_finite(tmp_path):
                         n_epochs=6, n_hidden=33, n_layers=3)
     for m in multi:
         assert np.isfinite(m["loss_history"]).all()
+
+
+def test_run_with_dist_eval_inductive(tmp_path):
+    """Inductive dist eval (dedicated train∪val / full-graph stores) must
+    match the thread-mode rank-0 Evaluator's semantics: both runs report
+    sane accuracies and the final dist val == Evaluator val on the same
+    final checkpoint."""
+    import torch
+    from bnsgcn_amd.models.models import create_model
+    from bnsgcn_amd.runtime.trainer import Evaluator
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        multi = _run_config(tmp_path, 2, model="graphsage", use_pp=True,
+                            sampling_rate=1.0, n_epochs=40, log_every=40,
+                            eval=True, eval_mode="dist", inductive=True,
+                            lr=0.05, n_hidden=32)
+        assert "test_acc" in multi[0]
+        name = "tiny-2-metis-vol-induc"
+        ck = f"checkpoint/{name}_final.pth.tar"
+        assert os.path.exists(ck)
+        # cross-check against the rank-0 CPU evaluator on the same weights
+        args = make_args(tmp_path, n_partitions=2, model="graphsage",
+                         use_pp=True, inductive=True, n_hidden=32)
+        ev = Evaluator(args)
+        m = create_model(args, n_feat=16, n_class=7, train_size=1)
+        m.load_state_dict(torch.load(ck))
+        want_val = ev.evaluate(m)["val"]
+        want_test = ev.evaluate_test(m)
+        assert abs(multi[0]["val_acc"] - want_val) < 1e-5
+        assert abs(multi[0]["test_acc"] - want_test) < 1e-5
+    finally:
+        os.chdir(cwd)
