@@ -345,3 +345,37 @@ def test_attn_project_matches_torch():
     torch.testing.assert_close(el, (z * al).sum(-1))
     torch.testing.assert_close(er, (z * ar).sum(-1))
     assert gradcheck(lambda a, b, c: F.attn_project(a, b, c), (z, al, ar))
+
+
+def test_parser_snake_case_aliases_and_defaults():
+    """Reference parser compatibility: kebab AND snake aliases accepted
+    (helper/parser.py), reference defaults preserved."""
+    from bnsgcn_amd.runtime.config import create_parser
+    p = create_parser()
+    a = p.parse_args(["--sampling_rate", "0.25", "--n_partitions", "4",
+                      "--n_hidden", "64", "--use_pp", "--fix_seed",
+                      "--partition_method", "random", "--weight_decay", "1e-4",
+                      "--part_path", "/tmp/pp", "--data_path", "/tmp/dd"])
+    assert a.sampling_rate == 0.25 and a.n_partitions == 4
+    assert a.n_hidden == 64 and a.use_pp and a.fix_seed
+    assert a.partition_method == "random" and a.weight_decay == 1e-4
+    assert a.partition_dir == "/tmp/pp" and a.data_path == "/tmp/dd"
+    d = p.parse_args([])
+    # reference defaults (helper/parser.py)
+    assert d.sampling_rate == 1.0 and d.lr == 1e-2 and d.n_epochs == 200
+    assert d.n_hidden == 16 and d.n_layers == 2 and d.eval is True
+    assert d.norm == "layer" and d.port == 18118
+
+
+def test_comm_timer_cpu_spans():
+    import time
+    from bnsgcn_amd.utils.timer import CommTimer
+    t = CommTimer()
+    with t.span("a"):
+        time.sleep(0.01)
+    with t.span("a"):
+        time.sleep(0.01)
+    tot = t.tot_time()
+    assert 0.015 < tot < 0.5
+    t.clear()
+    assert t.tot_time() == 0.0
