@@ -37,7 +37,7 @@ import torch.distributed as dist
 def parse():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--dataset", type=str, default="reddit")
     p.add_argument("--model", type=str, default="graphsage")

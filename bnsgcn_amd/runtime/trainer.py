@@ -379,13 +379,24 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
             pt = load_partition(args.partition_dir, name + "-evaltest", rank)
             eval_states = (EvalState(pv, args, device),
                            EvalState(pt, args, device))
+            have_stores = 1.0
         except FileNotFoundError:
+            have_stores = 0.0
+        # agree across ranks: if the stores are missing on ANY rank, every
+        # rank must fall back together, else dist_evaluate's collectives
+        # deadlock against thread-mode ranks
+        if world > 1:
+            flag = torch.tensor([have_stores])
+            dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+            have_stores = float(flag[0])
+        if have_stores < 1.0:
             if rank == 0:
                 print("inductive dist eval stores missing (re-run "
                       "partitioning with --eval-mode dist); falling back "
                       "to thread mode", flush=True)
             eval_mode = "thread"
             dist_eval_on = False
+            eval_states = None
     if args.eval and rank == 0:
         if not dist_eval_on:
             evaluator = Evaluator(args)

@@ -31,6 +31,18 @@ def main():
         prepare_partitions(args)
 
     world = args.n_partitions
+    if args.backend == "mpi":
+        # reference main.py:51-62: exec `mpirun -n P python train.py ...`;
+        # train.py discovers its rank from OMPI_COMM_WORLD_RANK
+        import shutil
+        import subprocess
+        import sys
+        if shutil.which("mpirun") is None:
+            raise RuntimeError("--backend mpi requires mpirun on PATH "
+                               "(reference main.py:51-62); use nccl/gloo")
+        cmd = ["mpirun", "-n", str(world), sys.executable, "train.py",
+               "--skip-partition"] + sys.argv[1:]
+        raise SystemExit(subprocess.call(cmd))
     start = args.node_rank * args.parts_per_node
     local = min(args.parts_per_node, world - start)
     os.environ.setdefault("MASTER_ADDR", args.master_addr)
