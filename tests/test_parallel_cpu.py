@@ -124,6 +124,60 @@ def _agg_rank(rank, world, parts_data, rate, n_epochs, mode="mean"):
     return float(err)
 
 
+def _reducer_rank(rank, world, overlap, bucket_bytes):
+    """Train a tiny MLP (one branch unused some steps) for 4 steps; return
+    the full grad arena after each synchronize. With overlap=True the
+    buckets launch from post-accumulate-grad hooks DURING backward; the
+    reduced values must be bit-identical to the post-backward path
+    (reference train.py:337-338 hook overlap)."""
+    import torch
+    from bnsgcn_amd.parallel import init_distributed
+    from bnsgcn_amd.parallel.reducer import GradReducer
+    init_distributed("gloo", rank, world)
+    torch.manual_seed(7)
+
+    class Net(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = torch.nn.Linear(17, 33)   # odd sizes: params straddle
+            self.b = torch.nn.Linear(33, 9)    # tiny-bucket boundaries
+            self.unused = torch.nn.Linear(5, 5)
+
+        def forward(self, x, use_extra):
+            h = torch.relu(self.a(x))
+            out = self.b(h)
+            if use_extra:
+                out = out + self.unused(out[:, :5]).sum() * 0
+            return out
+
+    net = Net()
+    red = GradReducer(net, n_train_global=10, bucket_bytes=bucket_bytes,
+                      overlap=overlap)
+    arenas = []
+    for step in range(4):
+        torch.manual_seed(100 + step)       # same data on both ranks except
+        x = torch.randn(6, 17) + rank       # the rank shift
+        red.zero_grad()
+        loss = net(x, use_extra=(step % 2 == 0)).pow(2).sum()
+        loss.backward()
+        red.reduce()
+        red.synchronize()
+        arenas.append(red.flat.detach().numpy().copy())
+    return arenas
+
+
+@pytest.mark.parametrize("bucket_bytes", [128, 1 << 20])
+def test_reducer_overlap_bit_identical(bucket_bytes):
+    a = run_dist(2, _reducer_rank, (True, bucket_bytes))
+    b = run_dist(2, _reducer_rank, (False, bucket_bytes))
+    for r in range(2):
+        for sa, sb in zip(a[r], b[r]):
+            np.testing.assert_array_equal(sa, sb)
+    # grads identical across ranks after all-reduce
+    for sa, sb in zip(a[0], a[1]):
+        np.testing.assert_array_equal(sa, sb)
+
+
 @pytest.mark.parametrize("mode", ["mean", "gcn"])
 def test_bns_estimator_unbiased(mode):
     from bnsgcn_amd.graph import load_data, partition_graph
