@@ -73,7 +73,16 @@ def all_to_all_rows(recv: torch.Tensor, send: torch.Tensor,
                                       input_split_sizes=send_counts,
                                       async_op=async_op)
     # gloo fallback: non-blocking pairwise isend/irecv, ring-ordered.
+    # gloo p2p is CPU-only; CUDA tensors are staged through host copies
+    # (the reference's gloo+GPU oversubscription mode, main.py:45 /
+    # utils.py:197-211 — minus the persistent pinned mirrors: this path
+    # is a test/compat fallback, the production path is RCCL above).
     rank, size = dist.get_rank(), dist.get_world_size()
+    staged = recv.is_cuda or send.is_cuda
+    d_recv, d_send = recv, send
+    if staged:
+        send = send.cpu()
+        recv = torch.empty_like(d_recv, device="cpu")
     s_off = [0]
     for c in send_counts:
         s_off.append(s_off[-1] + c)
@@ -94,6 +103,8 @@ def all_to_all_rows(recv: torch.Tensor, send: torch.Tensor,
     # self block
     if size >= 1 and recv_counts[rank] > 0:
         recv[r_off[rank]:r_off[rank + 1]].copy_(send[s_off[rank]:s_off[rank + 1]])
+    if staged:
+        d_recv.copy_(recv)
     return None
 
 
@@ -105,6 +116,9 @@ def exchange_counts(my_counts: torch.Tensor) -> torch.Tensor:
         return my_counts.clone()
     size = dist.get_world_size()
     rank = dist.get_rank()
+    dev = my_counts.device
+    if dist.get_backend() == "gloo" and my_counts.is_cuda:
+        my_counts = my_counts.cpu()   # gloo all_gather is CPU-only
     gathered = [torch.zeros_like(my_counts) for _ in range(size)]
     dist.all_gather(gathered, my_counts)
-    return torch.stack(gathered)[:, rank]
+    return torch.stack(gathered)[:, rank].to(dev)
