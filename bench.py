@@ -50,6 +50,8 @@ def parse():
     p.add_argument("--data-scale", type=float, default=1.0)
     p.add_argument("--partition-dir", type=str, default="bench_partition")
     p.add_argument("--device", type=str, default="auto")
+    p.add_argument("--backend", type=str, default="auto",
+                   choices=["auto", "nccl", "gloo"])
     return p.parse_args()
 
 
@@ -95,7 +97,9 @@ def main():
     if cuda:
         torch.cuda.set_device(torch.device(device))
     if world > 1:
-        rank, world = init_distributed("nccl" if cuda else "gloo")
+        backend = a.backend if a.backend != "auto" else \
+            ("nccl" if cuda else "gloo")
+        rank, world = init_distributed(backend)
     if rank == 0:
         args.skip_partition = True   # reuse an existing store for this config
         prepare_partitions(args)

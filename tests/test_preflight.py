@@ -117,7 +117,8 @@ def test_bench_multirank_gloo_cuda_oversubscribed():
     oversubscription, main.py:45)."""
     if not torch.cuda.is_available():
         pytest.skip("needs a GPU")
-    res = _run_bench_subprocess(2, extra=["--sampling-rate", "0.5"],
+    res = _run_bench_subprocess(2, extra=["--sampling-rate", "0.5",
+                                          "--backend", "gloo"],
                                 device="cuda:0")
     assert res["n_gpus"] == 2
     assert np.isfinite(res["value"]) and res["value"] > 0
