@@ -115,6 +115,9 @@ def main():
                          n_class=int(meta["n_class"]),
                          train_size=int(meta["n_train"])).to(device)
     state.precompute()
+    state.raw_feat = None   # bench never evaluates; frees the raw feature
+    if cuda:                # matrix (57 GB for 1-partition papers100M)
+        torch.cuda.empty_cache()
     if world > 1:
         for prm in model.parameters():
             dist.broadcast(prm.data, src=0)

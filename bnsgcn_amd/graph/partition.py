@@ -369,6 +369,31 @@ def partition_graph(g: Graph, n_parts: int, method: str = "metis", seed: int = 0
                     objective: str = "vol") -> tuple[list[Partition], dict]:
     """Split `g` into per-rank Partition objects (see store.Partition)."""
     n = g.n_nodes
+    meta = {
+        "n_parts": n_parts, "method": method, "objective": objective, "seed": seed,
+        "n_nodes": int(g.n_nodes), "n_edges": int(g.n_edges),
+        "n_feat": int(g.n_feat), "n_class": int(g.n_class),
+        "n_train": int(g.n_train), "multilabel": bool(g.multilabel),
+        "dataset": g.name,
+    }
+    if n_parts == 1:
+        # fast path: the whole graph IS the partition — no edge-level
+        # temporaries (papers100M: 1.7B-edge masks/gathers would dominate)
+        e0 = np.zeros(0, dtype=np.int32)
+        p0 = Partition(
+            rank=0, n_parts=1,
+            inner_global_nid=np.arange(n, dtype=np.int64),
+            feat=g.feat, label=g.label, train_mask=g.train_mask,
+            val_mask=g.val_mask, test_mask=g.test_mask,
+            in_deg=g.in_deg.astype(np.int32),
+            out_deg=g.out_deg.astype(np.int32),
+            inner_indptr=g.adj_in.indptr, inner_indices=g.adj_in.indices,
+            halo_part=e0, halo_owner_local=e0, halo_out_deg=e0,
+            halo_in_deg=e0,
+            halo_indptr=np.zeros(1, dtype=np.int64), halo_indices=e0,
+        )
+        p0.boundary = [np.zeros(0, dtype=np.int32)]
+        return [p0], meta
     part = assign_parts(n, n_parts, method, seed, adj=g.adj_in)
 
     # inner-local id of every node within its partition (sorted-global order)
@@ -453,14 +478,6 @@ def partition_graph(g: Graph, n_parts: int, method: str = "metis", seed: int = 0
             else np.zeros(0, dtype=np.int32)
             for j in range(n_parts)]
         parts[p].boundary[p] = np.zeros(0, dtype=np.int32)
-
-    meta = {
-        "n_parts": n_parts, "method": method, "objective": objective, "seed": seed,
-        "n_nodes": int(g.n_nodes), "n_edges": int(g.n_edges),
-        "n_feat": int(g.n_feat), "n_class": int(g.n_class),
-        "n_train": int(g.n_train), "multilabel": bool(g.multilabel),
-        "dataset": g.name,
-    }
     return parts, meta
 
 

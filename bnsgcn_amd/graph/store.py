@@ -76,8 +76,16 @@ class Partition:
 
 
 def save_partitions(parts: list[Partition], meta: dict, out_dir: str, graph_name: str) -> str:
+    from .synthetic import LazyFeat
     d = os.path.join(out_dir, graph_name)
     os.makedirs(d, exist_ok=True)
+    if parts and isinstance(parts[0].feat, LazyFeat):
+        # procedural features (papers100M scale): the store carries only
+        # the generator seed — partitions rebuild LazyFeat from their
+        # inner_global_nid at load and materialize on the training device
+        meta = dict(meta)
+        meta["procedural_feat"] = {"seed": parts[0].feat.seed,
+                                   "n_feat": parts[0].feat.shape[1]}
     with open(os.path.join(d, "meta.json"), "w") as f:
         json.dump(meta, f, indent=1)
     for p in parts:
@@ -86,6 +94,8 @@ def save_partitions(parts: list[Partition], meta: dict, out_dir: str, graph_name
             "in_deg", "out_deg", "inner_indptr", "inner_indices",
             "halo_part", "halo_owner_local", "halo_out_deg", "halo_in_deg",
             "halo_indptr", "halo_indices")}
+        if isinstance(p.feat, LazyFeat):
+            arrs["feat"] = np.zeros((0, p.feat.shape[1]), dtype=np.float32)
         for j, b in enumerate(p.boundary):
             arrs[f"boundary_{j}"] = b
         np.savez(os.path.join(d, f"part{p.rank}.npz"), **arrs)
@@ -107,4 +117,9 @@ def load_partition(part_dir: str, graph_name: str, rank: int) -> Partition:
         "in_deg", "out_deg", "inner_indptr", "inner_indices",
         "halo_part", "halo_owner_local", "halo_out_deg", "halo_in_deg",
         "halo_indptr", "halo_indices")}
+    pf = meta.get("procedural_feat")
+    if pf:
+        from .synthetic import LazyFeat
+        kw["feat"] = LazyFeat(int(pf["seed"]), int(pf["n_feat"]),
+                              kw["inner_global_nid"])
     return Partition(rank=rank, n_parts=n_parts, boundary=boundary, meta=meta, **kw)

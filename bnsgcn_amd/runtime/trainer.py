@@ -46,21 +46,34 @@ def _load_data_cached(args):
     cache = os.path.join(cache_dir,
                          f"_edges_{args.dataset}_s{args.seed}"
                          f"_x{args.data_scale:g}.npz")
+    from ..graph.synthetic import LazyFeat
     if os.path.exists(cache):
         z = np.load(cache)
-        adj = CSR(z["indptr"], z["indices"], int(z["n_nodes"]))
-        g = Graph(adj, z["feat"], z["label"], z["train_mask"], z["val_mask"],
+        n_nodes = int(z["n_nodes"])
+        adj = CSR(z["indptr"], z["indices"], n_nodes)
+        if "procedural_seed" in z:
+            feat = LazyFeat(int(z["procedural_seed"]), int(z["procedural_nf"]),
+                            np.arange(n_nodes, dtype=np.int64))
+        else:
+            feat = z["feat"]
+        g = Graph(adj, feat, z["label"], z["train_mask"], z["val_mask"],
                   z["test_mask"], int(z["n_class"]), bool(z["multilabel"]),
                   name=args.dataset)
     else:
         g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
+        extra = {}
+        if isinstance(g.feat, LazyFeat):
+            extra = {"feat": np.zeros((0, g.n_feat), dtype=np.float32),
+                     "procedural_seed": g.feat.seed, "procedural_nf": g.n_feat}
+        else:
+            extra = {"feat": g.feat}
         try:
             os.makedirs(cache_dir, exist_ok=True)
             np.savez(cache, indptr=g.adj_in.indptr, indices=g.adj_in.indices,
-                     n_nodes=g.n_nodes, feat=g.feat, label=g.label,
+                     n_nodes=g.n_nodes, label=g.label,
                      train_mask=g.train_mask, val_mask=g.val_mask,
                      test_mask=g.test_mask, n_class=g.n_class,
-                     multilabel=g.multilabel)
+                     multilabel=g.multilabel, **extra)
         except OSError:
             pass
     return g
@@ -100,6 +113,16 @@ def prepare_partitions(args) -> str:
 
 # ------------------------------------------------------------- rank setup
 
+def _feat_to_device(feat, device) -> torch.Tensor:
+    """numpy features -> device copy; LazyFeat (papers100M procedural
+    features) -> materialize directly into device memory (the 1-partition
+    papers matrix is 57 GB — it never exists on the host)."""
+    from ..graph.synthetic import LazyFeat
+    if isinstance(feat, LazyFeat):
+        return feat.materialize_torch(device)
+    return torch.from_numpy(feat).to(device)
+
+
 class RankState:
     """Everything one rank needs for the epoch loop."""
 
@@ -108,7 +131,7 @@ class RankState:
         self.part = part
         self.args = args
         dev = self.device
-        self.feat = torch.from_numpy(part.feat).to(dev)
+        self.feat = _feat_to_device(part.feat, dev)
         lab = torch.from_numpy(part.label)
         self.label = lab.to(dev)
         self.train_mask = torch.from_numpy(part.train_mask).to(dev)
@@ -164,7 +187,7 @@ class EvalState:
         self.device = torch.device(device)
         self.part = part
         self.args = args
-        self.feat = torch.from_numpy(part.feat).to(device)
+        self.feat = _feat_to_device(part.feat, device)
         self.raw_feat = self.feat
         self.label = torch.from_numpy(part.label).to(device)
         self.val_mask = torch.from_numpy(part.val_mask).to(device)
@@ -268,7 +291,7 @@ class Evaluator:
         ctx = GraphContext.for_full_graph(
             torch.from_numpy(g.adj_in.indptr), torch.from_numpy(g.adj_in.indices),
             torch.from_numpy(g.in_deg), torch.from_numpy(g.out_deg), self.device)
-        feat = torch.from_numpy(g.feat).to(self.device)
+        feat = _feat_to_device(g.feat, self.device)
         label = torch.from_numpy(g.label).to(self.device)
         return ctx, feat, label, g
 
