@@ -52,11 +52,16 @@ def parse():
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--backend", type=str, default="auto",
                    choices=["auto", "nccl", "gloo"])
+    p.add_argument("--use-pp", action=argparse.BooleanOptionalAction,
+                   default=True)
     return p.parse_args()
 
 
 def main():
     a = parse()
+    if a.dataset == "ogbn-papers100M" and a.data_scale >= 0.5:
+        # 111M-node tensors leave little slack for allocator fragmentation
+        os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
     from bnsgcn_amd.runtime.config import create_parser, graph_name_of
     from bnsgcn_amd.runtime.trainer import prepare_partitions, RankState, _forward
     from bnsgcn_amd.graph import load_partition, load_meta
@@ -79,7 +84,7 @@ def main():
     args.partition_method = a.partition_method
     args.data_scale = a.data_scale
     args.partition_dir = a.partition_dir
-    args.use_pp = True
+    args.use_pp = a.use_pp
     args.eval = False
     args.fix_seed = True
     args.seed = 0
@@ -114,10 +119,11 @@ def main():
     model = create_model(args, n_feat=int(meta["n_feat"]),
                          n_class=int(meta["n_class"]),
                          train_size=int(meta["n_train"])).to(device)
-    state.precompute()
-    state.raw_feat = None   # bench never evaluates; frees the raw feature
-    if cuda:                # matrix (57 GB for 1-partition papers100M)
-        torch.cuda.empty_cache()
+    if a.use_pp or a.model == "gat":
+        state.precompute()
+        state.raw_feat = None   # bench never evaluates; frees the raw
+        if cuda:                # features (57 GB for 1-partition papers100M)
+            torch.cuda.empty_cache()
     if world > 1:
         for prm in model.parameters():
             dist.broadcast(prm.data, src=0)
@@ -136,7 +142,10 @@ def main():
         state.plan.set_epoch(epoch)
         model.train()
         logits = _forward(model, state, state.feat)
-        loss = loss_fcn(logits[state.train_mask], labels_train)
+        if state.loss_rows is not None:     # final layer already restricted
+            loss = loss_fcn(logits, labels_train)
+        else:
+            loss = loss_fcn(logits[state.train_mask], labels_train)
         reducer.zero_grad()
         loss.backward()
         reducer.reduce()
@@ -192,7 +201,7 @@ def main():
                        "n_layers": a.n_layers, "n_hidden": a.n_hidden,
                        "sampling_rate": a.sampling_rate,
                        "n_nodes": int(meta.get("full_n_nodes", meta["n_nodes"])),
-                       "n_edges": n_edges, "use_pp": True,
+                       "n_edges": n_edges, "use_pp": a.use_pp,
                        "partition": a.partition_method,
                        "parallelism": f"partition-parallel p{world}",
                        "epoch_time_s": epoch_s},

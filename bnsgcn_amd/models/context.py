@@ -29,7 +29,7 @@ def _inv(x: torch.Tensor) -> torch.Tensor:
 
 class GraphContext:
     def __init__(self, indptr, indices, in_deg, out_deg, device,
-                 plan: HaloPlan | None = None):
+                 plan: HaloPlan | None = None, need_eperm: bool = True):
         dev = torch.device(device)
         self.plan = plan
         self.indptr = indptr.to(dev)
@@ -39,8 +39,19 @@ class GraphContext:
                                           if plan is None else plan.n_inner)
         self.t_indptr, self.t_indices = tip, tix
         # t_eperm[j] = original edge id of the j-th transposed edge (for
-        # per-edge payloads through the transpose — GAT attention weights)
-        self.t_eperm = t_eperm
+        # per-edge payloads through the transpose — GAT attention weights
+        # only; int64 per edge = 13.8 GB on papers100M, dropped otherwise)
+        self.t_eperm = t_eperm if need_eperm else None
+        # training-loss row restriction (set by the runtime): when model
+        # forward passes rows=loss_rows for the FINAL layer, aggregation
+        # and logits are computed only for labeled rows — identical math
+        # (unlabeled logits have zero gradient; only GEMM reduction-order
+        # float noise differs), and the
+        # [N, n_class] logits tensor shrinks to [n_train_local, n_class]
+        # (papers100M: 76 GB -> 0.8 GB)
+        self.loss_rows: torch.Tensor | None = None
+        self._rows_static: tuple | None = None   # gathered inner CSRs
+        self._rows_halo: tuple | None = None     # per-epoch gathered halo
         in_deg = in_deg.to(dev).float()
         out_deg = out_deg.to(dev).float()
         self.in_norm_inv = _inv_sqrt(in_deg)       # GCN dst scale
@@ -60,11 +71,13 @@ class GraphContext:
 
     # ---------------------------------------------------------------- train
     @classmethod
-    def for_partition(cls, part: Partition, plan: HaloPlan, device) -> "GraphContext":
+    def for_partition(cls, part: Partition, plan: HaloPlan, device,
+                      need_eperm: bool = True) -> "GraphContext":
         return cls(torch.from_numpy(part.inner_indptr),
                    torch.from_numpy(part.inner_indices),
                    torch.from_numpy(part.in_deg),
-                   torch.from_numpy(part.out_deg), device, plan=plan)
+                   torch.from_numpy(part.out_deg), device, plan=plan,
+                   need_eperm=need_eperm)
 
     # ----------------------------------------------------------------- eval
     @classmethod
@@ -76,22 +89,78 @@ class GraphContext:
     def inner_csrs(self):
         return (self.indptr, self.indices, self.t_indptr, self.t_indices)
 
-    def aggregate(self, x: torch.Tensor, mode: str) -> torch.Tensor:
+    def aggregate(self, x: torch.Tensor, mode: str,
+                  rows: torch.Tensor | None = None) -> torch.Tensor:
         """mode='gcn': symmetric-normalized sum (reference layer.py:32-38);
         mode='mean': in-degree mean with FULL-graph degrees
         (reference layer.py:85-92 — degrees precomputed before
         partitioning, utils.py:92-93, so the sampled estimator stays
-        unbiased after the 1/ratio pack scale)."""
+        unbiased after the 1/ratio pack scale).
+
+        rows: optional destination-row restriction (final-layer loss rows):
+        output is [len(rows), F], aggregating only into those rows. The
+        halo exchange itself is unchanged (senders sample independently of
+        receiver labels); only the local aggregation shrinks."""
         if mode == "gcn":
             src, dst, halo_src = self.out_norm_inv, self.in_norm_inv, True
         elif mode == "mean":
             src, dst, halo_src = None, self.in_deg_inv, False
         else:
             raise ValueError(mode)
+        if rows is not None:
+            dst = dst[rows]
         if self.plan is None:
             from ..ops.functional import spmm_sum
-            return spmm_sum(x, *self.inner_csrs, src_scale=src, dst_scale=dst)
-        return partition_aggregate(x, self.plan, self.inner_csrs, src, dst, halo_src)
+            if rows is None:
+                return spmm_sum(x, *self.inner_csrs, src_scale=src,
+                                dst_scale=dst)
+            ip, ix, tip, tix = self._rows_inner_csrs(rows)
+            return spmm_sum(x, ip, ix, tip, tix, src_scale=src, dst_scale=dst)
+        if rows is None:
+            return partition_aggregate(x, self.plan, self.inner_csrs, src,
+                                       dst, halo_src)
+        import dataclasses
+        st = self.plan.state
+        hfip, hfix, hbip, hbix = self._rows_halo_csrs(rows, st)
+        st_r = dataclasses.replace(st, halo_fwd_indptr=hfip,
+                                   halo_fwd_indices=hfix,
+                                   halo_bwd_indptr=hbip,
+                                   halo_bwd_indices=hbix)
+        ip, ix, tip, tix = self._rows_inner_csrs(rows)
+        from ..parallel.halo import _PartitionAggregate
+        return _PartitionAggregate.apply(x, self.plan, st_r, ip, ix, tip, tix,
+                                         src, dst, halo_src)
+
+    def _rows_inner_csrs(self, rows: torch.Tensor):
+        """Row-gathered inner CSR + transpose, cached (rows are static)."""
+        from ..ops.csr_torch import gather_rows_csr
+        if self._rows_static is None:
+            ip, ix = gather_rows_csr(self.indptr, self.indices, rows)
+            n_src = self.n_rows if self.plan is None else self.plan.n_inner
+            tip, tix, _ = transpose_csr(ip, ix, n_src)
+            if self.indptr.is_cuda:
+                from ..ops.functional import _worklist_of
+                _worklist_of(ip)
+                _worklist_of(tip)
+            self._rows_static = (ip, ix, tip, tix)
+        return self._rows_static
+
+    def _rows_halo_csrs(self, rows: torch.Tensor, st: EpochState):
+        """Row-gathered sampled-halo CSR + transpose, cached per epoch."""
+        from ..ops.csr_torch import gather_rows_csr
+        if self._rows_halo is not None and self._rows_halo[0] == st.epoch:
+            return self._rows_halo[1]
+        hfip, hfix = gather_rows_csr(st.halo_fwd_indptr, st.halo_fwd_indices,
+                                     rows)
+        R = int(st.halo_bwd_indptr.numel() - 1)
+        hbip, hbix, _ = transpose_csr(hfip, hfix, R)
+        if hfip.is_cuda:
+            from ..ops.functional import _worklist_of
+            _worklist_of(hfip)
+            _worklist_of(hbip)
+        out = (hfip, hfix, hbip, hbix)
+        self._rows_halo = (st.epoch, out)
+        return out
 
     # ------------------------------------------------------------- GAT block
     def gat_block(self):

@@ -45,10 +45,14 @@ class GCNLayer(nn.Module):
         if self.linear.bias is not None:
             _uniform_init(self.linear.bias, fan_in=in_feats)
 
-    def forward(self, ctx: GraphContext, x):
+    def forward(self, ctx: GraphContext, x, rows: torch.Tensor | None = None):
+        """rows: final-layer loss-row restriction — output [len(rows), F]
+        (identical math for those rows; see GraphContext.aggregate)."""
         if self.training and self.use_pp:
+            if rows is not None:
+                x = x[rows]
             return F.linear(x, self.linear.weight, self.linear.bias)
-        h = ctx.aggregate(x, "gcn")
+        h = ctx.aggregate(x, "gcn", rows=rows)
         return F.linear(h, self.linear.weight, self.linear.bias)
 
 
@@ -73,15 +77,18 @@ class SAGELayer(nn.Module):
                 _uniform_init(self.linear1.bias, self.linear2.bias,
                               fan_in=in_feats)
 
-    def forward(self, ctx: GraphContext, x):
+    def forward(self, ctx: GraphContext, x, rows: torch.Tensor | None = None):
         if self.training and self.use_pp:
             # x = [feat ‖ precomputed neighbor mean], width 2F
+            if rows is not None:
+                x = x[rows]
             return F.linear(x, self.linear.weight, self.linear.bias)
-        ah = ctx.aggregate(x, "mean")
+        ah = ctx.aggregate(x, "mean", rows=rows)
+        x_dst = x[rows] if rows is not None else x
         if self.use_pp:  # eval path of a pp layer (reference layer.py:98-100)
-            return F.linear(torch.cat((x, ah), dim=1), self.linear.weight,
+            return F.linear(torch.cat((x_dst, ah), dim=1), self.linear.weight,
                             self.linear.bias)
-        return (F.linear(x, self.linear1.weight, self.linear1.bias)
+        return (F.linear(x_dst, self.linear1.weight, self.linear1.bias)
                 + F.linear(ah, self.linear2.weight, self.linear2.bias))
 
 

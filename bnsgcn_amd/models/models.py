@@ -63,13 +63,21 @@ class GCN(GNNBase):
             pp = False
 
     def forward(self, ctx: GraphContext, feat):
+        """When training and ctx.loss_rows is set, the FINAL layer computes
+        only the loss rows (logits shape [len(loss_rows), C]) — identical
+        math, since unlabeled rows carry zero gradient (differences are
+        GEMM reduction-order float noise). Dropout stays full-shaped so
+        RNG draws match the unrestricted run."""
         h = feat
+        restrict = self.training and getattr(ctx, "loss_rows", None) is not None
         for i in range(self.n_layers):
+            rows = ctx.loss_rows if (restrict and i == self.n_layers - 1) \
+                else None
             h = self.dropout(h)
             if i < self.n_layers - self.n_linear:
-                h = self.layers[i](ctx, h)
+                h = self.layers[i](ctx, h, rows=rows)
             else:
-                h = self.layers[i](h)
+                h = self.layers[i](h[rows] if rows is not None else h)
             h = self._post(i, h)
         return h
 

@@ -141,10 +141,20 @@ class RankState:
         self.plan = HaloPlan(part, args.sampling_rate, seed=args.seed, device=dev,
                              unit_ratio=(args.model == "gat"
                                          and not args.gat_ratio_scale))
-        self.ctx = GraphContext.for_partition(part, self.plan, dev)
+        self.ctx = GraphContext.for_partition(part, self.plan, dev,
+                                              need_eperm=(args.model == "gat"))
         self.halo_feat0 = None     # GAT use_pp layer-0 full halo features
         self.n_train_global = int(part.meta["n_train"])
         self.part_train = int(part.train_mask.sum())
+        # final-layer loss-row restriction (GCN/SAGE): train-time logits are
+        # computed only for labeled rows — identical loss/gradients, and
+        # [N, C] logits shrink to [n_train_local, C] (papers100M: 76 GB ->
+        # 0.8 GB). BNSGCN_FULL_LOGITS=1 restores the reference's full pass.
+        self.loss_rows = None
+        if (os.environ.get("BNSGCN_FULL_LOGITS") != "1"
+                and args.model != "gat"):
+            self.loss_rows = torch.nonzero(self.train_mask).flatten()
+            self.ctx.loss_rows = self.loss_rows
 
     # ---------------------------------------------------------- precompute
     @torch.no_grad()
@@ -253,6 +263,15 @@ def _forward(model, state: RankState, feat):
     if state.args.model == "gat":
         return model(state.ctx, feat, halo_feat0=state.halo_feat0)
     return model(state.ctx, feat)
+
+
+def forward_train_logits(model, state: RankState) -> torch.Tensor:
+    """Training forward returning logits FOR THE TRAIN ROWS, regardless of
+    whether the final-layer loss-row restriction is active."""
+    logits = _forward(model, state, state.feat)
+    if getattr(state, "loss_rows", None) is not None and model.training:
+        return logits
+    return logits[state.train_mask]
 
 
 # -------------------------------------------------------------- evaluation
@@ -450,7 +469,10 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
         state.plan.set_epoch(epoch)
         model.train()
         logits = _forward(model, state, state.feat)
-        loss = loss_fcn(logits[state.train_mask], labels_train)
+        if state.loss_rows is not None:     # final layer already restricted
+            loss = loss_fcn(logits, labels_train)
+        else:
+            loss = loss_fcn(logits[state.train_mask], labels_train)
         reducer.zero_grad()
         loss.backward()
         t_red = time.perf_counter()

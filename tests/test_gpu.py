@@ -144,7 +144,7 @@ def test_single_rank_training_step_gpu(model):
     from bnsgcn_amd.graph import load_data, partition_graph
     from bnsgcn_amd.models.models import create_model
     from bnsgcn_amd.runtime.config import create_parser
-    from bnsgcn_amd.runtime.trainer import RankState, _forward
+    from bnsgcn_amd.runtime.trainer import RankState, _forward, forward_train_logits
     from bnsgcn_amd.parallel import GradReducer
 
     args = create_parser().parse_args([])
@@ -174,8 +174,8 @@ def test_single_rank_training_step_gpu(model):
     for ep in range(15):
         state.plan.set_epoch(ep)
         m.train()
-        logits = _forward(m, state, state.feat)
-        loss = lf(logits[state.train_mask], state.label[state.train_mask].long())
+        logits = forward_train_logits(m, state)
+        loss = lf(logits, state.label[state.train_mask].long())
         reducer.zero_grad()
         loss.backward()
         reducer.synchronize()
@@ -192,7 +192,7 @@ def test_gpu_matches_cpu_training():
     from bnsgcn_amd.graph import load_data, partition_graph
     from bnsgcn_amd.models.models import create_model
     from bnsgcn_amd.runtime.config import create_parser
-    from bnsgcn_amd.runtime.trainer import RankState, _forward
+    from bnsgcn_amd.runtime.trainer import RankState, _forward, forward_train_logits
     from bnsgcn_amd.parallel import GradReducer
 
     def train(device):
@@ -219,9 +219,8 @@ def test_gpu_matches_cpu_training():
         losses = []
         for ep in range(10):
             m.train()
-            logits = _forward(m, state, state.feat)
-            loss = lf(logits[state.train_mask],
-                      state.label[state.train_mask].long())
+            logits = forward_train_logits(m, state)
+            loss = lf(logits, state.label[state.train_mask].long())
             reducer.zero_grad()
             loss.backward()
             reducer.synchronize()
@@ -301,7 +300,7 @@ def test_train_and_eval_on_gpu_accuracy():
     from bnsgcn_amd.graph import load_data, partition_graph
     from bnsgcn_amd.models.models import create_model
     from bnsgcn_amd.runtime.config import create_parser
-    from bnsgcn_amd.runtime.trainer import RankState, _forward, Evaluator
+    from bnsgcn_amd.runtime.trainer import RankState, _forward, forward_train_logits, forward_train_logits, Evaluator
     from bnsgcn_amd.parallel import GradReducer
 
     args = create_parser().parse_args([])
@@ -329,8 +328,8 @@ def test_train_and_eval_on_gpu_accuracy():
     lf = torch.nn.CrossEntropyLoss(reduction="sum")
     for ep in range(300):
         m.train()
-        logits = _forward(m, state, state.feat)
-        loss = lf(logits[state.train_mask], state.label[state.train_mask].long())
+        logits = forward_train_logits(m, state)
+        loss = lf(logits, state.label[state.train_mask].long())
         reducer.zero_grad()
         loss.backward()
         reducer.synchronize()

@@ -267,8 +267,10 @@ def test_big_graph_training_step(tmp_path, monkeypatch):
                          n_class=int(part.meta["n_class"]),
                          train_size=int(part.meta["n_train"]))
     logits = model(state.ctx, state.feat)
+    # training mode restricts the final layer to loss rows by default
+    assert logits.shape[0] == int(state.train_mask.sum())
     loss = torch.nn.functional.cross_entropy(
-        logits[state.train_mask], state.label[state.train_mask].long())
+        logits, state.label[state.train_mask].long())
     loss.backward()
     assert torch.isfinite(loss)
 

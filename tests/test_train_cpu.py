@@ -76,6 +76,35 @@ def test_dist_matches_single_p4(tmp_path):
     np.testing.assert_allclose(lh_multi, lh_single, rtol=5e-3, atol=1e-3)
 
 
+@pytest.mark.parametrize("model,n_linear,world,rate",
+                         [("graphsage", 0, 1, 1.0), ("gcn", 0, 1, 1.0),
+                          ("graphsage", 1, 1, 1.0),
+                          ("graphsage", 0, 2, 0.5), ("gcn", 1, 2, 0.5)])
+def test_loss_row_restriction_bit_identical(tmp_path, model, n_linear,
+                                            world, rate):
+    """Final-layer loss-row restriction (default on) must reproduce the
+    full-logits loss trajectory to float reduction-order noise — same
+    math, same dropout RNG draws; the only difference is GEMM reduction
+    shape ([R,*] vs [N,*] with zero rows), worth ~1e-7 relative per epoch
+    — single-process and world=2, sampled and full rate (this is what
+    makes papers100M trainable at 1 GPU). Epoch 0 (pre-update) is exact."""
+    import os
+    kw = dict(model=model, n_linear=n_linear, sampling_rate=rate,
+              use_pp=(n_linear == 0), n_epochs=6, dropout=0.3)
+    os.environ.pop("BNSGCN_FULL_LOGITS", None)
+    on = _run_config(tmp_path / "on", world, **kw)
+    os.environ["BNSGCN_FULL_LOGITS"] = "1"
+    try:
+        off = _run_config(tmp_path / "off", world, **kw)
+    finally:
+        os.environ.pop("BNSGCN_FULL_LOGITS", None)
+    for r in range(world):
+        a = np.array(on[r]["loss_history"])
+        b = np.array(off[r]["loss_history"])
+        assert a[0] == b[0]                      # pre-update epoch: exact
+        np.testing.assert_allclose(a, b, rtol=1e-4, atol=1e-4)
+
+
 @pytest.mark.parametrize("model", ["graphsage", "gcn"])
 def test_sampled_training_learns(tmp_path, model):
     multi = _run_config(tmp_path, 2, model=model, sampling_rate=0.3,
@@ -288,7 +317,8 @@ def _dist_eval_check(rank, world, args):
     full-graph Evaluator on the SAME weights — must agree exactly."""
     import torch
     from bnsgcn_amd.runtime.trainer import (run, dist_evaluate, Evaluator,
-                                            RankState, _forward)
+                                            RankState, _forward,
+                                            forward_train_logits)
     from bnsgcn_amd.graph import load_partition
     from bnsgcn_amd.models.models import create_model
     from bnsgcn_amd.parallel import init_distributed, GradReducer
@@ -312,7 +342,7 @@ def _dist_eval_check(rank, world, args):
     for ep in range(30):
         state.plan.set_epoch(ep)
         model.train()
-        loss = lf(_forward(model, state, state.feat)[state.train_mask],
+        loss = lf(forward_train_logits(model, state),
                   state.label[state.train_mask].long())
         red.zero_grad()
         loss.backward()

@@ -49,7 +49,7 @@ def main():
     trainer.all_to_all_rows = fake_all_to_all_rows
 
     from bnsgcn_amd.runtime.config import create_parser, graph_name_of
-    from bnsgcn_amd.runtime.trainer import prepare_partitions, RankState, _forward
+    from bnsgcn_amd.runtime.trainer import prepare_partitions, RankState, _forward, forward_train_logits
     from bnsgcn_amd.graph import load_partition
     from bnsgcn_amd.models.models import create_model
     from bnsgcn_amd.parallel import GradReducer
@@ -85,7 +85,7 @@ def main():
     def step(e):
         state.plan.set_epoch(e)
         model.train()
-        loss = lf(_forward(model, state, state.feat)[state.train_mask], labels)
+        loss = lf(forward_train_logits(model, state), labels)
         reducer.zero_grad()
         loss.backward()
         reducer.reduce()
