@@ -54,6 +54,7 @@ def parse():
                    choices=["auto", "nccl", "gloo"])
     p.add_argument("--use-pp", action=argparse.BooleanOptionalAction,
                    default=True)
+    p.add_argument("--dropout", type=float, default=0.5)
     return p.parse_args()
 
 
@@ -85,6 +86,7 @@ def main():
     args.data_scale = a.data_scale
     args.partition_dir = a.partition_dir
     args.use_pp = a.use_pp
+    args.dropout = a.dropout
     args.eval = False
     args.fix_seed = True
     args.seed = 0
