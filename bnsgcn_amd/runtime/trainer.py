@@ -42,7 +42,13 @@ def _load_data_cached(args):
     scaling sweep) skip the expensive edge sampling."""
     import numpy as np
     from ..graph.csr import CSR, Graph
-    cache_dir = getattr(args, "data_path", None) or args.partition_dir
+    from ..graph.ingest import disk_dataset_file, load_disk_data
+    data_path = getattr(args, "data_path", None)
+    if disk_dataset_file(args.dataset, data_path):
+        # pre-downloaded real dataset on disk (reference helper/utils.py
+        # loads via dgl/ogb + network; see graph/ingest.py for the layout)
+        return load_disk_data(args.dataset, data_path)
+    cache_dir = data_path or args.partition_dir
     cache = os.path.join(cache_dir,
                          f"_edges_{args.dataset}_s{args.seed}"
                          f"_x{args.data_scale:g}.npz")
@@ -297,7 +303,12 @@ class Evaluator:
         self.args = args
         self.device = torch.device(getattr(args, "eval_device", "cpu"))
         self.multilabel = None
-        g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
+        from ..graph.ingest import disk_dataset_file, load_disk_data
+        dp = getattr(args, "data_path", None)
+        if disk_dataset_file(args.dataset, dp):
+            g = load_disk_data(args.dataset, dp)
+        else:
+            g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
         self.multilabel = g.multilabel
         self._graphs = {}
         if args.inductive:

