@@ -151,6 +151,9 @@ def main():
         reducer.zero_grad()
         loss.backward()
         reducer.reduce()
+        # build next epoch's sampling plan on a side stream, overlapped
+        # with this epoch's queued GPU work
+        state.prefetch(epoch + 1)
         reducer.synchronize()
         optimizer.step()
         return loss

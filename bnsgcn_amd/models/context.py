@@ -146,10 +146,27 @@ class GraphContext:
         return self._rows_static
 
     def _rows_halo_csrs(self, rows: torch.Tensor, st: EpochState):
-        """Row-gathered sampled-halo CSR + transpose, cached per epoch."""
-        from ..ops.csr_torch import gather_rows_csr
+        """Row-gathered sampled-halo CSR + transpose, cached per epoch
+        (optionally pre-built on a side stream by prefetch_rows_halo)."""
         if self._rows_halo is not None and self._rows_halo[0] == st.epoch:
+            ev = self._rows_halo[2]
+            if ev is not None:      # built on the prefetch stream
+                cur = torch.cuda.current_stream()
+                cur.wait_event(ev)
+                for t in self._rows_halo[1]:
+                    t.record_stream(cur)
+                    wl = getattr(t, "_bns_worklist", None)
+                    if wl is not None:
+                        for w in wl:
+                            w.record_stream(cur)
+                self._rows_halo = (st.epoch, self._rows_halo[1], None)
             return self._rows_halo[1]
+        out = self._build_rows_halo(rows, st)
+        self._rows_halo = (st.epoch, out, None)
+        return out
+
+    def _build_rows_halo(self, rows: torch.Tensor, st: EpochState):
+        from ..ops.csr_torch import gather_rows_csr
         hfip, hfix = gather_rows_csr(st.halo_fwd_indptr, st.halo_fwd_indices,
                                      rows)
         R = int(st.halo_bwd_indptr.numel() - 1)
@@ -158,9 +175,18 @@ class GraphContext:
             from ..ops.functional import _worklist_of
             _worklist_of(hfip)
             _worklist_of(hbip)
-        out = (hfip, hfix, hbip, hbix)
-        self._rows_halo = (st.epoch, out)
-        return out
+        return (hfip, hfix, hbip, hbix)
+
+    def prefetch_rows_halo(self, st: EpochState, stream) -> None:
+        """Build the restricted-halo CSRs for a FUTURE epoch state on the
+        given side stream (called by RankState.prefetch alongside
+        HaloPlan.prefetch)."""
+        if self.loss_rows is None:
+            return
+        with torch.cuda.stream(stream):
+            out = self._build_rows_halo(self.loss_rows, st)
+            ev = stream.record_event()
+        self._rows_halo = (st.epoch, out, ev)
 
     # ------------------------------------------------------------- GAT block
     def gat_block(self):

@@ -49,6 +49,22 @@ class EpochState:
     halo_in_deg: torch.Tensor = None         # [R] full in-degree of recv rows
 
 
+def _state_tensors(st: EpochState):
+    """Every device tensor in an EpochState, incl. cached worklists (for
+    cross-stream record_stream when the state was built by prefetch)."""
+    out = []
+    for f in ("pack_idx", "pack_scale", "hsel", "halo_fwd_indptr",
+              "halo_fwd_indices", "halo_bwd_indptr", "halo_bwd_indices",
+              "halo_eperm_t", "halo_out_norm_inv", "halo_in_deg"):
+        t = getattr(st, f)
+        if t is not None:
+            out.append(t)
+            wl = getattr(t, "_bns_worklist", None)
+            if wl is not None:
+                out.extend(wl)
+    return out
+
+
 class HaloPlan:
     """One per training job per rank. `set_epoch(e)` refreshes sampling."""
 
@@ -105,9 +121,46 @@ class HaloPlan:
             sample_boundary(n, s, self.seed, epoch, src, dst)).to(dev)
 
     # ------------------------------------------------------------------
+    def prefetch(self, epoch: int) -> None:
+        """Build epoch `epoch`'s sampling state on a SIDE stream so the
+        gathers/sorts/transposes overlap the CURRENT epoch's backward on
+        the GPU (the plan is a pure function of seed+epoch — independent
+        of training state). `set_epoch(epoch)` then just installs it.
+        No-op on CPU or for static rates."""
+        import os
+        if self._static or self.device.type != "cuda" \
+                or os.environ.get("BNSGCN_NO_PREFETCH") == "1":
+            return
+        import torch as _t
+        if not hasattr(self, "_prefetch_stream"):
+            self._prefetch_stream = _t.cuda.Stream()
+        s = self._prefetch_stream
+        ev = _t.cuda.current_stream().record_event()
+        with _t.cuda.stream(s):
+            s.wait_event(ev)        # boundary/halo tensors are settled
+            st = self._build(epoch)
+            done = s.record_event()
+        self._next = (epoch, st, done)
+
     def set_epoch(self, epoch: int) -> EpochState:
         if self._static and self._state is not None:
             return self._state
+        nxt = getattr(self, "_next", None)
+        if nxt is not None and nxt[0] == epoch:
+            self._next = None
+            _, st, done = nxt
+            import torch as _t
+            cur = _t.cuda.current_stream()
+            cur.wait_event(done)
+            for t in _state_tensors(st):
+                t.record_stream(cur)
+            self._state = st
+            return st
+        st = self._build(epoch)
+        self._state = st
+        return st
+
+    def _build(self, epoch: int) -> EpochState:
         dev = self.device
         me = self.rank
         pack_parts, scale_parts, hsel_parts = [], [], []
@@ -158,7 +211,6 @@ class HaloPlan:
         )
         st.send_counts[me] = 0
         st.recv_counts[me] = 0
-        self._state = st
         return st
 
     @property

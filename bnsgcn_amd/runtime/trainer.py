@@ -162,6 +162,15 @@ class RankState:
             self.loss_rows = torch.nonzero(self.train_mask).flatten()
             self.ctx.loss_rows = self.loss_rows
 
+    def prefetch(self, epoch: int) -> None:
+        """Overlap the NEXT epoch's sampling-plan build (and the
+        restricted-halo gather when loss-row restriction is on) with the
+        current epoch's queued GPU work, on a side stream."""
+        self.plan.prefetch(epoch)
+        nxt = getattr(self.plan, "_next", None)
+        if nxt is not None and self.ctx.loss_rows is not None:
+            self.ctx.prefetch_rows_halo(nxt[1], self.plan._prefetch_stream)
+
     # ---------------------------------------------------------- precompute
     @torch.no_grad()
     def precompute(self):
@@ -488,6 +497,7 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
         loss.backward()
         t_red = time.perf_counter()
         reducer.reduce()
+        state.prefetch(epoch + 1)        # overlap next epoch's sampling
         reducer.synchronize()
         reduce_dur.append(time.perf_counter() - t_red)
         optimizer.step()

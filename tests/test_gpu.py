@@ -186,6 +186,59 @@ def test_single_rank_training_step_gpu(model):
 
 
 @needs_gpu
+def test_plan_prefetch_trajectory_identical():
+    """Side-stream plan prefetch (RankState.prefetch) must be trajectory-
+    identical to building the sampling plan on the main stream — same
+    Philox draws, only the stream changes."""
+    from bnsgcn_amd.graph import load_data, partition_graph
+    from bnsgcn_amd.models.models import create_model
+    from bnsgcn_amd.runtime.config import create_parser
+    from bnsgcn_amd.runtime.trainer import RankState, forward_train_logits
+    from bnsgcn_amd.parallel import GradReducer
+
+    def train(prefetch: bool):
+        args = create_parser().parse_args([])
+        args.dataset = "tiny"
+        args.model = "graphsage"
+        args.n_layers = 3
+        args.n_hidden = 16
+        args.sampling_rate = 0.5
+        args.use_pp = True
+        args.dropout = 0.0
+        torch.manual_seed(0)
+        g = load_data("tiny", seed=0)
+        parts, meta = partition_graph(g, 1, method="metis")
+        parts[0].meta = meta
+        state = RankState(parts[0], args, "cuda:0")
+        state.plan.set_epoch(0)
+        m = create_model(args, n_feat=g.n_feat, n_class=g.n_class,
+                         train_size=g.n_train).to("cuda:0")
+        state.precompute()
+        reducer = GradReducer(m, g.n_train)
+        opt = torch.optim.Adam(m.parameters(), lr=1e-2)
+        lf = torch.nn.CrossEntropyLoss(reduction="sum")
+        losses = []
+        for ep in range(12):
+            state.plan.set_epoch(ep)
+            m.train()
+            logits = forward_train_logits(m, state)
+            loss = lf(logits, state.label[state.train_mask].long())
+            reducer.zero_grad()
+            loss.backward()
+            if prefetch:
+                state.prefetch(ep + 1)
+            reducer.synchronize()
+            opt.step()
+            losses.append(loss.item())
+        torch.cuda.synchronize()
+        return np.array(losses)
+
+    a = train(prefetch=True)
+    b = train(prefetch=False)
+    np.testing.assert_array_equal(a, b)
+
+
+@needs_gpu
 def test_gpu_matches_cpu_training():
     """GPU single-rank loss trajectory ≈ CPU single-rank (same seed, no
     dropout): validates the whole HIP op set against the torch reference."""
