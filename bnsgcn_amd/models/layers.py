@@ -33,6 +33,23 @@ def _uniform_init(*tensors, fan_in):
         nn.init.uniform_(t, -stdv, stdv)
 
 
+class LayerNorm(nn.Module):
+    """Drop-in affine nn.LayerNorm over the last dim: HIP ln_fwd/ln_bwd
+    kernels on GPU (K7), torch on CPU. Same parameter names as
+    nn.LayerNorm so `norm.<i>.weight/bias` state-dict keys (and reference
+    checkpoints) stay compatible."""
+
+    def __init__(self, dim: int, elementwise_affine: bool = True, eps=1e-5):
+        super().__init__()
+        assert elementwise_affine
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return F.layer_norm(x, self.weight, self.bias, self.eps)
+
+
 class GCNLayer(nn.Module):
     """State-dict keys match the reference exactly (layers.<i>.linear.*,
     module/layer.py:17) so checkpoints are interchangeable."""

@@ -13,7 +13,7 @@ from __future__ import annotations
 from torch import nn
 
 from .context import GraphContext
-from .layers import GCNLayer, SAGELayer, GATLayer
+from .layers import GCNLayer, SAGELayer, GATLayer, LayerNorm
 from .sync_bn import SyncBatchNorm
 
 
@@ -31,8 +31,8 @@ class GNNBase(nn.Module):
             self.norm = nn.ModuleList()
             for i in range(self.n_layers - 1):
                 if norm == "layer":
-                    self.norm.append(nn.LayerNorm(layer_size[i + 1],
-                                                  elementwise_affine=True))
+                    self.norm.append(LayerNorm(layer_size[i + 1],
+                                               elementwise_affine=True))
                 else:
                     self.norm.append(SyncBatchNorm(layer_size[i + 1], train_size))
         self.dropout = nn.Dropout(p=dropout)

@@ -277,6 +277,31 @@ def linear(x, weight, bias=None):
     return _Linear.apply(x, weight, bias)
 
 
+class _LayerNormF(Function):
+    """Row LayerNorm (K7) on the fused gfx950 kernels — torch's ROCm LN
+    kernels measured ~27% of the ogbn-products epoch (profiles/
+    topk_products_r02.txt); this pair runs at HBM streaming rate."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        x = x.contiguous()
+        y, mean, rstd = get_ext().ln_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, mean, rstd = ctx.saved_tensors
+        dx, dw, db = get_ext().ln_bwd(x, dy.contiguous(), w, mean, rstd)
+        return dx, dw, db, None
+
+
+def layer_norm(x, weight, bias, eps: float = 1e-5):
+    if use_hip(x) and x.shape[-1] <= 1024 and x.dtype == torch.float32:
+        return _LayerNormF.apply(x, weight, bias, eps)
+    return torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
 def segment_softmax2_raw(indptr1, logits1, indptr2, logits2):
     if use_hip(logits1):
         return get_ext().segment_softmax2(indptr1, logits1, indptr2, logits2)

@@ -859,6 +859,102 @@ __global__ void attn_project_dattn_kernel(
   else { atomicAdd(&dal[f], sl); atomicAdd(&dar[f], sr); }
 }
 
+// ============================ LayerNorm (K7) ===========================
+// Row-wise LayerNorm over [N, F] fp32 (F <= 1024). One wave per row,
+// 4 rows per 256-thread block, grid-stride; row values stay in registers
+// across both reduction passes, lane l covers columns l, l+64, ...
+// (coalesced). Replaces torch's ROCm LN kernels, which measured ~27% of
+// the ogbn-products epoch (profiles/topk_products_r02.txt: 1 TB/s
+// effective vs ~6 TB/s for this shape).
+
+#define LN_MAX_K 16  // supports F up to 64*16
+
+__global__ __launch_bounds__(256) void ln_fwd_kernel(
+    const float* __restrict__ x, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ y,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    int64_t n, int F, float eps) {
+  const int lane = threadIdx.x & 63;
+  const float inv_f = 1.0f / F;
+  float vals[LN_MAX_K], gm[LN_MAX_K], bt[LN_MAX_K];
+  int K = 0;
+  for (int f = lane; f < F; f += WAVE, ++K) { gm[K] = gamma[f]; bt[K] = beta[f]; }
+  const int64_t stride = (int64_t)gridDim.x * 4;
+  for (int64_t r = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6); r < n;
+       r += stride) {
+    const float* xr = x + r * (int64_t)F;
+    float s = 0.f;
+    int k = 0;
+    for (int f = lane; f < F; f += WAVE, ++k) { vals[k] = xr[f]; s += vals[k]; }
+    s = wave_reduce_sum(s);
+    const float mu = s * inv_f;
+    float v = 0.f;
+    for (int i = 0; i < K; ++i) { const float d = vals[i] - mu; v += d * d; }
+    v = wave_reduce_sum(v);
+    const float rstd = rsqrtf(v * inv_f + eps);
+    float* yr = y + r * (int64_t)F;
+    k = 0;
+    for (int f = lane; f < F; f += WAVE, ++k)
+      yr[f] = (vals[k] - mu) * rstd * gm[k] + bt[k];
+    if (lane == 0) { mean_out[r] = mu; rstd_out[r] = rstd; }
+  }
+}
+
+// dx = rstd * ( g - mean(g) - xhat * mean(g * xhat) ),  g = dy * gamma
+__global__ __launch_bounds__(256) void ln_bwd_dx_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    const float* __restrict__ gamma, const float* __restrict__ mean,
+    const float* __restrict__ rstd, float* __restrict__ dx,
+    int64_t n, int F) {
+  const int lane = threadIdx.x & 63;
+  const float inv_f = 1.0f / F;
+  float xh[LN_MAX_K], g[LN_MAX_K], gm[LN_MAX_K];
+  int K = 0;
+  for (int f = lane; f < F; f += WAVE, ++K) gm[K] = gamma[f];
+  const int64_t stride = (int64_t)gridDim.x * 4;
+  for (int64_t r = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6); r < n;
+       r += stride) {
+    const float* xr = x + r * (int64_t)F;
+    const float* dr = dy + r * (int64_t)F;
+    const float mu = mean[r], rs = rstd[r];
+    float a = 0.f, b = 0.f;
+    int k = 0;
+    for (int f = lane; f < F; f += WAVE, ++k) {
+      xh[k] = (xr[f] - mu) * rs;
+      g[k] = dr[f] * gm[k];
+      a += g[k];
+      b += g[k] * xh[k];
+    }
+    a = wave_reduce_sum(a) * inv_f;
+    b = wave_reduce_sum(b) * inv_f;
+    float* dxr = dx + r * (int64_t)F;
+    k = 0;
+    for (int f = lane; f < F; f += WAVE, ++k)
+      dxr[f] = rs * (g[k] - a - xh[k] * b);
+  }
+}
+
+// dgamma[f] = sum_r dy*xhat, dbeta[f] = sum_r dy — column reduction in
+// the syncbn_stats shape (thread per column, chunked rows, atomic combine)
+__global__ void ln_bwd_w_kernel(
+    const float* __restrict__ x, const float* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    float* __restrict__ out, int64_t n, int F, int row_chunks) {
+  const int f = blockIdx.x * blockDim.x + threadIdx.x;
+  if (f >= F) return;
+  const int64_t rows_per = (n + row_chunks - 1) / row_chunks;
+  const int64_t r0 = (int64_t)blockIdx.y * rows_per;
+  const int64_t r1 = (r0 + rows_per < n) ? r0 + rows_per : n;
+  float sg = 0.f, sb = 0.f;
+  for (int64_t r = r0; r < r1; ++r) {
+    const float d = dy[r * (int64_t)F + f];
+    sg += d * (x[r * (int64_t)F + f] - mean[r]) * rstd[r];
+    sb += d;
+  }
+  if (row_chunks == 1) { out[f] = sg; out[F + f] = sb; }
+  else { atomicAdd(&out[f], sg); atomicAdd(&out[F + f], sb); }
+}
+
 // ------------------------------ launchers ------------------------------
 
 inline void check_f32(const at::Tensor& t, const char* name) {
@@ -993,6 +1089,50 @@ at::Tensor syncbn_stats(at::Tensor x) {
                      x.data_ptr<float>(), out.data_ptr<float>(), n, F,
                      row_chunks);
   return out;
+}
+
+std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor gamma,
+                               at::Tensor beta, double eps) {
+  check_f32(x, "x");
+  check_f32(gamma, "gamma");
+  check_f32(beta, "beta");
+  const int64_t n = x.size(0);
+  const int F = (int)x.size(1);
+  TORCH_CHECK(F <= 64 * LN_MAX_K, "ln_fwd: F too large (", F, ")");
+  auto y = at::empty_like(x);
+  auto mean = at::empty({n}, x.options());
+  auto rstd = at::empty({n}, x.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int blocks = (int)std::min<int64_t>((n + 3) / 4, 32768);
+  hipLaunchKernelGGL(ln_fwd_kernel, dim3(std::max(blocks, 1)), dim3(256), 0,
+                     stream, x.data_ptr<float>(), gamma.data_ptr<float>(),
+                     beta.data_ptr<float>(), y.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), n, F,
+                     (float)eps);
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
+                               at::Tensor mean, at::Tensor rstd) {
+  check_f32(x, "x");
+  check_f32(dy, "dy");
+  const int64_t n = x.size(0);
+  const int F = (int)x.size(1);
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({2, F}, x.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int blocks = (int)std::min<int64_t>((n + 3) / 4, 32768);
+  hipLaunchKernelGGL(ln_bwd_dx_kernel, dim3(std::max(blocks, 1)), dim3(256),
+                     0, stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), dx.data_ptr<float>(), n, F);
+  const int row_chunks = (int)std::min<int64_t>((n + 255) / 256, 2048);
+  dim3 grid((F + 255) / 256, row_chunks);
+  hipLaunchKernelGGL(ln_bwd_w_kernel, grid, dim3(256), 0, stream,
+                     x.data_ptr<float>(), dy.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     dw.data_ptr<float>(), n, F, row_chunks);
+  return {dx, dw[0], dw[1]};
 }
 
 at::Tensor bincount_i32(at::Tensor v, int64_t n_bins) {
@@ -1297,6 +1437,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("spmm_edge_sum", &spmm_edge_sum, "multi-head edge-weighted SpMM");
   m.def("sddmm_dot", &sddmm_dot, "per-edge per-head dot (spmm_edge grad)");
   m.def("bincount_i32", &bincount_i32, "atomic int32 histogram");
+  m.def("ln_fwd", &ln_fwd, "row LayerNorm forward (y, mean, rstd)");
+  m.def("ln_bwd", &ln_bwd, "row LayerNorm backward (dx, dgamma, dbeta)");
   m.def("attn_project", &attn_project, "fused GAT el/er projections");
   m.def("attn_project_backward", &attn_project_backward,
         "fused GAT projection backward");

@@ -186,6 +186,32 @@ def test_single_rank_training_step_gpu(model):
 
 
 @needs_gpu
+@pytest.mark.parametrize("F", [41, 64, 128, 256, 600, 1024])
+def test_layernorm_matches_torch(F):
+    """HIP ln_fwd/ln_bwd (K7) vs torch LayerNorm: outputs and all three
+    gradients, including F not a multiple of the 64-lane wave."""
+    from bnsgcn_amd.ops.functional import layer_norm
+    torch.manual_seed(0)
+    n = 4097
+    x = (torch.randn(n, F, device="cuda:0") * 3 + 1).requires_grad_(True)
+    w = torch.randn(F, device="cuda:0").requires_grad_(True)
+    b = torch.randn(F, device="cuda:0").requires_grad_(True)
+    y = layer_norm(x, w, b)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = torch.nn.functional.layer_norm(x2, (F,), w2, b2, 1e-5)
+    y2.backward(g)
+    torch.testing.assert_close(y, y2, rtol=2e-5, atol=2e-5)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=2e-4, atol=2e-4)
+    torch.testing.assert_close(w.grad, w2.grad, rtol=2e-3, atol=2e-3)
+    torch.testing.assert_close(b.grad, b2.grad, rtol=2e-3, atol=2e-3)
+
+
+@needs_gpu
 def test_plan_prefetch_trajectory_identical():
     """Side-stream plan prefetch (RankState.prefetch) must be trajectory-
     identical to building the sampling plan on the main stream — same
