@@ -170,9 +170,10 @@ class GATLayer(nn.Module):
         li = F.sddmm_add(el_in, er, ctx.indptr, ctx.indices,
                          ctx.t_indptr, ctx.t_indices, ctx.t_eperm, slope=slope)
         lh = F.sddmm_add(el_h, er, hip, hix, hbip, hbix, heperm, slope=slope)
-        ai, ah = F.segment_softmax2(li, lh, ctx.indptr, hip)
-        ai = self.attn_drop(ai)
-        ah = self.attn_drop(ah)
+        # attention dropout FUSED into the union softmax (no separate
+        # dropout kernels/masks; identity when eval or p=0)
+        p_attn = self.attn_drop.p if self.training else 0.0
+        ai, ah = F.segment_softmax2(li, lh, ctx.indptr, hip, p_drop=p_attn)
         out = F.spmm_edge_sum(z_in, ai, ctx.indptr, ctx.indices,
                               ctx.t_indptr, ctx.t_indices, ctx.t_eperm)
         out = out + F.spmm_edge_sum(z_h, ah, hip, hix, hbip, hbix, heperm)

@@ -304,36 +304,66 @@ def layer_norm(x, weight, bias, eps: float = 1e-5):
 
 def segment_softmax2_raw(indptr1, logits1, indptr2, logits2):
     if use_hip(logits1):
-        return get_ext().segment_softmax2(indptr1, logits1, indptr2, logits2)
+        return get_ext().segment_softmax2(indptr1, logits1, indptr2, logits2,
+                                          1.0, 0)[:2]
     return ref.segment_softmax2(indptr1, logits1, indptr2, logits2)
 
 
 def segment_softmax2_bwd_raw(indptr1, a1, g1, indptr2, a2, g2):
     if use_hip(a1):
         return get_ext().segment_softmax2_backward(indptr1, a1, g1,
-                                                   indptr2, a2, g2)
+                                                   indptr2, a2, g2, 1.0, 0)
     return ref.segment_softmax2_backward(indptr1, a1, g1, indptr2, a2, g2)
 
 
 class _SegmentSoftmax2(Function):
-    """Union softmax over two per-row edge sets (split GAT block)."""
+    """Union softmax over two per-row edge sets (split GAT block), with
+    the GAT attention dropout FUSED on the HIP path (p_drop > 0): the
+    forward emits the dropped weights alongside the saved pre-drop
+    softmax, and the backward regenerates the Philox mask from the saved
+    seed — no mask tensor, no separate dropout kernels (reference: DGL
+    GATConv attn_drop)."""
 
     @staticmethod
-    def forward(ctx, logits1, logits2, indptr1, indptr2):
-        a1, a2 = segment_softmax2_raw(indptr1, logits1, indptr2, logits2)
+    def forward(ctx, logits1, logits2, indptr1, indptr2, p_drop):
+        keep = 1.0 - float(p_drop or 0.0)
+        if use_hip(logits1):
+            seed = int(torch.randint(0, 2**62, (1,)).item()) if keep < 1.0 \
+                else 0
+            a1, a2, da1, da2 = get_ext().segment_softmax2(
+                indptr1, logits1, indptr2, logits2, keep, seed)
+        else:
+            a1, a2 = ref.segment_softmax2(indptr1, logits1, indptr2, logits2)
+            seed = 0
+            if keep < 1.0:
+                m1 = (torch.rand_like(a1) < keep).float()
+                m2 = (torch.rand_like(a2) < keep).float()
+                da1, da2 = a1 * m1 / keep, a2 * m2 / keep
+                ctx.cpu_masks = (m1, m2)
+            else:
+                da1, da2 = a1, a2
+        ctx.keep, ctx.seed = keep, seed
         ctx.save_for_backward(a1, a2, indptr1, indptr2)
-        return a1, a2
+        return da1, da2
 
     @staticmethod
     def backward(ctx, g1, g2):
         a1, a2, indptr1, indptr2 = ctx.saved_tensors
-        d1, d2 = segment_softmax2_bwd_raw(indptr1, a1, g1.contiguous(),
-                                          indptr2, a2, g2.contiguous())
-        return d1, d2, None, None
+        g1, g2 = g1.contiguous(), g2.contiguous()
+        if use_hip(a1):
+            d1, d2 = get_ext().segment_softmax2_backward(
+                indptr1, a1, g1, indptr2, a2, g2, ctx.keep, ctx.seed)
+        else:
+            if ctx.keep < 1.0:
+                m1, m2 = ctx.cpu_masks
+                g1, g2 = g1 * m1 / ctx.keep, g2 * m2 / ctx.keep
+            d1, d2 = ref.segment_softmax2_backward(indptr1, a1, g1,
+                                                   indptr2, a2, g2)
+        return d1, d2, None, None, None
 
 
-def segment_softmax2(logits1, logits2, indptr1, indptr2):
-    return _SegmentSoftmax2.apply(logits1, logits2, indptr1, indptr2)
+def segment_softmax2(logits1, logits2, indptr1, indptr2, p_drop=0.0):
+    return _SegmentSoftmax2.apply(logits1, logits2, indptr1, indptr2, p_drop)
 
 
 class _AttnProject(Function):

@@ -481,16 +481,35 @@ __global__ void segment_softmax_bwd_kernel(const int64_t* __restrict__ indptr,
 
 // Union softmax over TWO per-row edge segments (split GAT block:
 // static inner edges + per-epoch sampled halo edges) — shared max/sum.
+// Dropout mask for fused attention dropout: element idx keeps its value
+// iff philox(c0=idx_lo, c1=idx_hi, key=seed)[0] < keep * 2^32. The mask
+// is REGENERATED in backward from (seed, idx) — never stored.
+DEV_INLINE bool drop_keep(int64_t idx, uint32_t thr, uint64_t seed) {
+  const P4 p = philox4x32((uint32_t)idx, (uint32_t)(idx >> 32), 0, 0,
+                          (uint32_t)seed, (uint32_t)(seed >> 32));
+  return p.c0 < thr;
+}
+
+// keep < 1: additionally writes the attn-dropout-applied weights into
+// da1/da2 (the spmm input), while a1/a2 keep the pre-drop softmax (the
+// backward state) — replaces the separate torch dropout pass
+// (reference: DGL GATConv attn_drop).
 __global__ void segment_softmax2_kernel(const int64_t* __restrict__ ip1,
                                         const float* __restrict__ l1,
                                         const int64_t* __restrict__ ip2,
                                         const float* __restrict__ l2,
                                         float* __restrict__ a1,
                                         float* __restrict__ a2,
-                                        int n_rows, int H) {
+                                        float* __restrict__ da1,
+                                        float* __restrict__ da2,
+                                        int n_rows, int H, float keep,
+                                        uint64_t seed, int64_t off2) {
   const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const bool drop = keep < 1.0f;
+  const uint32_t thr = (uint32_t)(keep * 4294967296.0);
+  const float inv_keep = drop ? 1.0f / keep : 1.0f;
   for (int rh = wave; rh < n_rows * H; rh += n_waves) {
     const int r = rh / H, h = rh % H;
     const int64_t b1 = ip1[r], e1 = ip1[r + 1];
@@ -505,13 +524,25 @@ __global__ void segment_softmax2_kernel(const int64_t* __restrict__ ip1,
     for (int64_t e = b2 + lane; e < e2; e += WAVE) sum += __expf(l2[e * H + h] - m);
     sum = wave_reduce_sum(sum);
     const float inv = 1.0f / fmaxf(sum, 1e-38f);
-    for (int64_t e = b1 + lane; e < e1; e += WAVE)
-      a1[e * H + h] = __expf(l1[e * H + h] - m) * inv;
-    for (int64_t e = b2 + lane; e < e2; e += WAVE)
-      a2[e * H + h] = __expf(l2[e * H + h] - m) * inv;
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) {
+      const float a = __expf(l1[e * H + h] - m) * inv;
+      a1[e * H + h] = a;
+      if (drop)
+        da1[e * H + h] = drop_keep(e * H + h, thr, seed) ? a * inv_keep : 0.f;
+    }
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) {
+      const float a = __expf(l2[e * H + h] - m) * inv;
+      a2[e * H + h] = a;
+      if (drop)
+        da2[e * H + h] = drop_keep(off2 + e * H + h, thr, seed)
+                             ? a * inv_keep : 0.f;
+    }
   }
 }
 
+// keep < 1: the incoming grads g1/g2 are w.r.t. the DROPPED weights; the
+// dropout backward (mask/keep, mask regenerated from seed) is folded in
+// before the softmax Jacobian.
 __global__ void segment_softmax2_bwd_kernel(const int64_t* __restrict__ ip1,
                                             const float* __restrict__ a1,
                                             const float* __restrict__ g1,
@@ -520,24 +551,42 @@ __global__ void segment_softmax2_bwd_kernel(const int64_t* __restrict__ ip1,
                                             const float* __restrict__ g2,
                                             float* __restrict__ d1,
                                             float* __restrict__ d2,
-                                            int n_rows, int H) {
+                                            int n_rows, int H, float keep,
+                                            uint64_t seed, int64_t off2) {
   const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const bool drop = keep < 1.0f;
+  const uint32_t thr = (uint32_t)(keep * 4294967296.0);
+  const float inv_keep = drop ? 1.0f / keep : 1.0f;
   for (int rh = wave; rh < n_rows * H; rh += n_waves) {
     const int r = rh / H, h = rh % H;
     const int64_t b1 = ip1[r], e1 = ip1[r + 1];
     const int64_t b2 = ip2[r], e2 = ip2[r + 1];
     float sum = 0.f;
-    for (int64_t e = b1 + lane; e < e1; e += WAVE)
-      sum += a1[e * H + h] * g1[e * H + h];
-    for (int64_t e = b2 + lane; e < e2; e += WAVE)
-      sum += a2[e * H + h] * g2[e * H + h];
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) {
+      float g = g1[e * H + h];
+      if (drop) g = drop_keep(e * H + h, thr, seed) ? g * inv_keep : 0.f;
+      sum += a1[e * H + h] * g;
+    }
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) {
+      float g = g2[e * H + h];
+      if (drop)
+        g = drop_keep(off2 + e * H + h, thr, seed) ? g * inv_keep : 0.f;
+      sum += a2[e * H + h] * g;
+    }
     sum = wave_reduce_sum(sum);
-    for (int64_t e = b1 + lane; e < e1; e += WAVE)
-      d1[e * H + h] = a1[e * H + h] * (g1[e * H + h] - sum);
-    for (int64_t e = b2 + lane; e < e2; e += WAVE)
-      d2[e * H + h] = a2[e * H + h] * (g2[e * H + h] - sum);
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) {
+      float g = g1[e * H + h];
+      if (drop) g = drop_keep(e * H + h, thr, seed) ? g * inv_keep : 0.f;
+      d1[e * H + h] = a1[e * H + h] * (g - sum);
+    }
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) {
+      float g = g2[e * H + h];
+      if (drop)
+        g = drop_keep(off2 + e * H + h, thr, seed) ? g * inv_keep : 0.f;
+      d2[e * H + h] = a2[e * H + h] * (g - sum);
+    }
   }
 }
 
@@ -666,6 +715,56 @@ __global__ __launch_bounds__(256) void spmm_edge_vec4_kernel(
         out4[o] = pv;
       } else {
         out4[o] = acc;
+      }
+    }
+  }
+}
+
+// sddmm_dot, vec4 general form (any H, D % 4 == 0): same 4x16-subgroup
+// edge pipeline as the gen kernel but float4 loads (4x fewer memory ops;
+// the D=100 GAT output layer measured 7.8 ms/call on the scalar gen
+// path, profiles/topk_gat_r02.txt)
+__global__ __launch_bounds__(256) void sddmm_dot_vec4_kernel(
+    const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
+    const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
+    const int32_t* __restrict__ indices, const float* __restrict__ g,
+    const float* __restrict__ x, float* __restrict__ out, int H, int D,
+    int HD) {
+  const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
+  const int wv = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int it_beg = wave_start[wv], it_end = wave_start[wv + 1];
+  const int sg = lane >> 4;            // 4 subgroups of 16 lanes
+  const int gl = lane & 15;
+  const int D4 = D >> 2;
+  const float4* __restrict__ x4 = reinterpret_cast<const float4*>(x);
+  const float4* __restrict__ g4 = reinterpret_cast<const float4*>(g);
+  for (int it = it_beg; it < it_end; ++it) {
+    int row = wrow[it];
+    if (row < 0) row = ~row;
+    const int64_t beg = wbeg[it], end = wend[it];
+    const int64_t grow = (int64_t)row * (HD >> 2);
+    for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
+      const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
+      int cid = 0;
+      if (lane < nv) cid = indices[e0 + lane];
+      for (int k0 = 0; k0 < nv; k0 += 4) {
+        const int k = k0 + sg;
+        const bool act = k < nv;
+        const int c = __shfl(cid, act ? k : 0, WAVE);
+        const int64_t crow = (int64_t)c * (HD >> 2);
+        for (int h = 0; h < H; ++h) {
+          float p = 0.f;
+          if (act)
+            for (int d = gl; d < D4; d += 16) {
+              const float4 a = g4[grow + h * D4 + d];
+              const float4 b = x4[crow + h * D4 + d];
+              p += a.x * b.x + a.y * b.y + a.z * b.z + a.w * b.w;
+            }
+#pragma unroll
+          for (int off = 8; off > 0; off >>= 1) p += __shfl_xor(p, off, WAVE);
+          if (act && gl == 0) out[(e0 + k) * H + h] = p;
+        }
       }
     }
   }
@@ -1193,25 +1292,32 @@ at::Tensor segment_softmax_backward(at::Tensor indptr, at::Tensor alpha,
 }
 
 std::vector<at::Tensor> segment_softmax2(at::Tensor ip1, at::Tensor l1,
-                                         at::Tensor ip2, at::Tensor l2) {
+                                         at::Tensor ip2, at::Tensor l2,
+                                         double keep, int64_t seed) {
   check_f32(l1, "l1"); check_f32(l2, "l2");
   const int n_rows = ip1.numel() - 1;
   const int H = l1.size(1);
+  const bool drop = keep < 1.0;
   auto a1 = at::empty_like(l1);
   auto a2 = at::empty_like(l2);
-  if (n_rows == 0) return {a1, a2};
+  auto da1 = drop ? at::empty_like(l1) : a1;
+  auto da2 = drop ? at::empty_like(l2) : a2;
+  if (n_rows == 0) return {a1, a2, da1, da2};
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(segment_softmax2_kernel, dim3(spmm_grid(n_rows * H)),
                      dim3(256), 0, stream, ip1.data_ptr<int64_t>(),
                      l1.data_ptr<float>(), ip2.data_ptr<int64_t>(),
                      l2.data_ptr<float>(), a1.data_ptr<float>(),
-                     a2.data_ptr<float>(), n_rows, H);
-  return {a1, a2};
+                     a2.data_ptr<float>(), da1.data_ptr<float>(),
+                     da2.data_ptr<float>(), n_rows, H, (float)keep,
+                     (uint64_t)seed, l1.numel());
+  return {a1, a2, da1, da2};
 }
 
 std::vector<at::Tensor> segment_softmax2_backward(at::Tensor ip1, at::Tensor a1,
                                                   at::Tensor g1, at::Tensor ip2,
-                                                  at::Tensor a2, at::Tensor g2) {
+                                                  at::Tensor a2, at::Tensor g2,
+                                                  double keep, int64_t seed) {
   const int n_rows = ip1.numel() - 1;
   const int H = a1.size(1);
   auto d1 = at::empty_like(a1);
@@ -1223,7 +1329,8 @@ std::vector<at::Tensor> segment_softmax2_backward(at::Tensor ip1, at::Tensor a1,
                      a1.data_ptr<float>(), g1.data_ptr<float>(),
                      ip2.data_ptr<int64_t>(), a2.data_ptr<float>(),
                      g2.data_ptr<float>(), d1.data_ptr<float>(),
-                     d2.data_ptr<float>(), n_rows, H);
+                     d2.data_ptr<float>(), n_rows, H, (float)keep,
+                     (uint64_t)seed, a1.numel());
   return {d1, d2};
 }
 
@@ -1281,6 +1388,15 @@ at::Tensor sddmm_dot(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                        wstart.data_ptr<int32_t>(), indices.data_ptr<int32_t>(),
                        g.data_ptr<float>(), x.data_ptr<float>(),
                        out.data_ptr<float>(), H, D, HD);
+    return out;
+  }
+  if (D % 4 == 0) {
+    hipLaunchKernelGGL(sddmm_dot_vec4_kernel, dim3(n_waves / 4), dim3(256),
+                       0, stream, wrow.data_ptr<int32_t>(),
+                       wbeg.data_ptr<int64_t>(), wend.data_ptr<int64_t>(),
+                       wstart.data_ptr<int32_t>(), indices.data_ptr<int32_t>(),
+                       g.data_ptr<float>(), x.data_ptr<float>(),
+                       out.data_ptr<float>(), H, D, H * D);
     return out;
   }
   hipLaunchKernelGGL(sddmm_dot_gen_kernel, dim3(n_waves / 4), dim3(256), 0,

@@ -186,6 +186,51 @@ def test_single_rank_training_step_gpu(model):
 
 
 @needs_gpu
+def test_fused_attn_dropout_softmax2():
+    """Fused attention dropout in segment_softmax2: dropped weights equal
+    a*mask/keep for the (regenerated-Philox) mask, the kept fraction is
+    ~keep, and the backward equals the manual dropout∘softmax chain
+    computed with the SAME mask."""
+    from bnsgcn_amd.ops import functional as BF
+    from bnsgcn_amd.ops import reference as ref
+    torch.manual_seed(0)
+    n, H = 500, 4
+    dev = "cuda:0"
+    deg1 = torch.randint(0, 6, (n,))
+    deg2 = torch.randint(0, 4, (n,))
+    ip1 = torch.zeros(n + 1, dtype=torch.long)
+    ip1[1:] = deg1.cumsum(0)
+    ip2 = torch.zeros(n + 1, dtype=torch.long)
+    ip2[1:] = deg2.cumsum(0)
+    ip1, ip2 = ip1.to(dev), ip2.to(dev)
+    l1 = torch.randn(int(deg1.sum()), H, device=dev, requires_grad=True)
+    l2 = torch.randn(int(deg2.sum()), H, device=dev, requires_grad=True)
+    keep = 0.7
+
+    da1, da2 = BF.segment_softmax2(l1, l2, ip1, ip2, p_drop=1 - keep)
+    a1, a2 = BF.segment_softmax2_raw(ip1, l1.detach(), ip2, l2.detach())
+    m1 = (da1 != 0).float()
+    m2 = (da2 != 0).float()
+    torch.testing.assert_close(da1, a1 * m1 / keep, rtol=1e-6, atol=1e-7)
+    torch.testing.assert_close(da2, a2 * m2 / keep, rtol=1e-6, atol=1e-7)
+    tot = m1.numel() + m2.numel()
+    frac = float(m1.sum() + m2.sum()) / tot
+    assert abs(frac - keep) < 0.03, frac
+
+    g1 = torch.randn_like(da1)
+    g2 = torch.randn_like(da2)
+    (da1 * g1).sum().backward(retain_graph=True)
+    (da2 * g2).sum().backward()
+
+    # manual chain with the same mask on the CPU reference
+    d1r, d2r = ref.segment_softmax2_backward(
+        ip1.cpu(), a1.cpu(), (g1 * m1 / keep).cpu(),
+        ip2.cpu(), a2.cpu(), (g2 * m2 / keep).cpu())
+    torch.testing.assert_close(l1.grad.cpu(), d1r, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(l2.grad.cpu(), d2r, rtol=1e-4, atol=1e-5)
+
+
+@needs_gpu
 @pytest.mark.parametrize("F", [41, 64, 128, 256, 600, 1024])
 def test_layernorm_matches_torch(F):
     """HIP ln_fwd/ln_bwd (K7) vs torch LayerNorm: outputs and all three
