@@ -174,9 +174,11 @@ class GATLayer(nn.Module):
         # dropout kernels/masks; identity when eval or p=0)
         p_attn = self.attn_drop.p if self.training else 0.0
         ai, ah = F.segment_softmax2(li, lh, ctx.indptr, hip, p_drop=p_attn)
-        out = F.spmm_edge_sum(z_in, ai, ctx.indptr, ctx.indices,
-                              ctx.t_indptr, ctx.t_indices, ctx.t_eperm)
-        out = out + F.spmm_edge_sum(z_h, ah, hip, hix, hbip, hbix, heperm)
+        out = F.spmm_edge_sum2(
+            z_in, ai, z_h, ah,
+            (ctx.indptr, ctx.indices, ctx.t_indptr, ctx.t_indices,
+             ctx.t_eperm),
+            (hip, hix, hbip, hbix, heperm))
         if self.bias is not None:
             out = out + self.bias.view(1, H, D)
         return out

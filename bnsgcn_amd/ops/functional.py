@@ -161,6 +161,45 @@ def spmm_edge_sum(x, w, indptr, indices, indptr_t, indices_t, eperm_t):
     return _SpMMEdge.apply(x, w, indptr, indices, indptr_t, indices_t, eperm_t)
 
 
+class _SpMMEdge2(Function):
+    """Inner + halo GAT aggregation accumulated into ONE output buffer —
+    removes the [N, H, D]-sized elementwise add between the two edge-set
+    results (~5% of the Yelp GAT epoch in bare CUDAFunctor_add kernels,
+    profiles/topk_gat2_r02.txt)."""
+
+    @staticmethod
+    def forward(ctx, x_in, w_in, x_h, w_h,
+                ip, ix, tip, tix, eperm_t,
+                hip_, hix_, hbip, hbix, heperm_t):
+        ctx.save_for_backward(x_in, w_in, x_h, w_h, ip, ix, tip, tix,
+                              eperm_t, hip_, hix_, hbip, hbix, heperm_t)
+        out = spmm_edge_raw(ip, ix, w_in, x_in)
+        spmm_edge_raw(hip_, hix_, w_h, x_h, out=out)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        (x_in, w_in, x_h, w_h, ip, ix, tip, tix, eperm_t,
+         hip_, hix_, hbip, hbix, heperm_t) = ctx.saved_tensors
+        grad = grad.contiguous()
+        gx_in = gw_in = gx_h = gw_h = None
+        if ctx.needs_input_grad[0]:
+            gx_in = spmm_edge_raw(tip, tix, w_in, grad, wperm=eperm_t)
+        if ctx.needs_input_grad[1]:
+            gw_in = sddmm_dot_raw(ip, ix, grad, x_in)
+        if ctx.needs_input_grad[2]:
+            gx_h = spmm_edge_raw(hbip, hbix, w_h, grad, wperm=heperm_t)
+        if ctx.needs_input_grad[3]:
+            gw_h = sddmm_dot_raw(hip_, hix_, grad, x_h)
+        return (gx_in, gw_in, gx_h, gw_h) + (None,) * 10
+
+
+def spmm_edge_sum2(x_in, w_in, x_h, w_h, inner_csrs5, halo_csrs5):
+    """inner_csrs5 = (ip, ix, tip, tix, eperm_t); halo_csrs5 likewise."""
+    return _SpMMEdge2.apply(x_in, w_in, x_h, w_h,
+                            *inner_csrs5, *halo_csrs5)
+
+
 class _SDDMMAdd(Function):
     """logits[e,h] = [leaky_relu](el[col_e,h] + er[row_e,h])  (u_add_v
     SDDMM, optionally fused with the GAT LeakyReLU — the activation is

@@ -482,12 +482,17 @@ __global__ void segment_softmax_bwd_kernel(const int64_t* __restrict__ indptr,
 // Union softmax over TWO per-row edge segments (split GAT block:
 // static inner edges + per-epoch sampled halo edges) — shared max/sum.
 // Dropout mask for fused attention dropout: element idx keeps its value
-// iff philox(c0=idx_lo, c1=idx_hi, key=seed)[0] < keep * 2^32. The mask
-// is REGENERATED in backward from (seed, idx) — never stored.
+// iff splitmix64(seed + idx) high bits < keep * 2^32. The mask is
+// REGENERATED in backward from (seed, idx) — never stored. splitmix64
+// (~8 ALU ops) instead of Philox4x32-10: the softmax kernel is
+// latency-bound over short segments and the 10-round Philox measured
+// +1.6 ms on the Yelp GAT backward (profiles/topk_gat2_r02.txt).
 DEV_INLINE bool drop_keep(int64_t idx, uint32_t thr, uint64_t seed) {
-  const P4 p = philox4x32((uint32_t)idx, (uint32_t)(idx >> 32), 0, 0,
-                          (uint32_t)seed, (uint32_t)(seed >> 32));
-  return p.c0 < thr;
+  uint64_t z = seed + (uint64_t)idx;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z ^= z >> 31;
+  return (uint32_t)(z >> 32) < thr;
 }
 
 // keep < 1: additionally writes the attn-dropout-applied weights into
