@@ -48,18 +48,34 @@ __global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
     const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
     const int32_t* __restrict__ indices, const float* __restrict__ x,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
-    float* __restrict__ out, int f4) {
+    float* __restrict__ out, int f4, int strided) {
   // SUBW = scheduling width: 64 (full wave) for wide rows, 32 (half-wave
   // subgroups, independent items per half) when f4 <= 32 so narrow
   // feature dims (e.g. h=128 -> f4=32) keep every lane busy.
   const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
   const int w = bb * (blockDim.x / SUBW) + (threadIdx.x / SUBW);
   const int lane = threadIdx.x & (SUBW - 1);
-  const int it_beg = wave_start[w], it_end = wave_start[w + 1];
+  // strided: the ~128 co-resident waves of an XCD walk CONSECUTIVE
+  // worklist items (stride = waves-per-XCD) instead of each owning a
+  // private contiguous range — one shared sliding src window in the
+  // XCD's 4 MB L2 rather than 128 disjoint ones (the r1 PMC profile
+  // showed 21.5% TCC hit = L2 thrash, profiles/pmc_spmm_micro_r01.txt).
+  int it_beg, it_end, it_step = 1;
+  if (strided) {
+    const int n_w = gridDim.x * (blockDim.x / SUBW);
+    const int wpx = n_w >> 3;
+    const int xcd = w / wpx, q = w - xcd * wpx;
+    it_beg = wave_start[xcd * wpx] + q;
+    it_end = wave_start[(xcd + 1) * wpx];
+    it_step = wpx;
+  } else {
+    it_beg = wave_start[w];
+    it_end = wave_start[w + 1];
+  }
   const float4* __restrict__ x4 = reinterpret_cast<const float4*>(x);
   float4* __restrict__ out4 = reinterpret_cast<float4*>(out);
 
-  for (int it = it_beg; it < it_end; ++it) {
+  for (int it = it_beg; it < it_end; it += it_step) {
     int row = wrow[it];
     const bool atomic = row < 0;
     if (atomic) row = ~row;
@@ -1097,8 +1113,13 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
   if (wrow.numel() == 0 || n_waves <= 0) return out;
   auto stream = at::cuda::getCurrentCUDAStream();
   const int grid = n_waves / 4;
+  static const int strided_env = [] {
+    const char* e = getenv("BNSGCN_SPMM_STRIDED");
+    return e ? atoi(e) : 1;                 // default: shared-window mode
+  }();
   if (F % 4 == 0) {
     if (F / 4 <= 32 && n_waves % 8 == 0) {
+      const int strided = strided_env && (n_waves % 8 == 0);
       auto kfn = acc ? spmm_sum_vec4_kernel<true, 32>
                      : spmm_sum_vec4_kernel<false, 32>;
       hipLaunchKernelGGL(kfn, dim3(n_waves / 8), dim3(256), 0, stream,
@@ -1106,9 +1127,10 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                          wend.data_ptr<int64_t>(), wave_start.data_ptr<int32_t>(),
                          indices.data_ptr<int32_t>(), x.data_ptr<float>(),
                          opt_ptr(src_scale), opt_ptr(dst_scale),
-                         out.data_ptr<float>(), F / 4);
+                         out.data_ptr<float>(), F / 4, strided);
       return out;
     }
+    const int strided = strided_env && (n_waves % 8 == 0);
     auto kfn = acc ? spmm_sum_vec4_kernel<true, 64>
                    : spmm_sum_vec4_kernel<false, 64>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
@@ -1116,7 +1138,7 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                        wend.data_ptr<int64_t>(), wave_start.data_ptr<int32_t>(),
                        indices.data_ptr<int32_t>(), x.data_ptr<float>(),
                        opt_ptr(src_scale), opt_ptr(dst_scale),
-                       out.data_ptr<float>(), F / 4);
+                       out.data_ptr<float>(), F / 4, strided);
   } else {
     auto kfn = acc ? spmm_sum_scalar_kernel<true> : spmm_sum_scalar_kernel<false>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
