@@ -1115,7 +1115,8 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
   const int grid = n_waves / 4;
   static const int strided_env = [] {
     const char* e = getenv("BNSGCN_SPMM_STRIDED");
-    return e ? atoi(e) : 1;                 // default: shared-window mode
+    return e ? atoi(e) : 0;  // measured: strided LOSES ~1-6%
+    // (drifting waves break their own temporal locality; bench_strided_*)
   }();
   if (F % 4 == 0) {
     if (F / 4 <= 32 && n_waves % 8 == 0) {
