@@ -278,14 +278,24 @@ def segment_softmax(logits, indptr):
     return _SegmentSoftmax.apply(logits, indptr)
 
 
+import os as _os
+
+# GEMM dispatch (K6). Per-shape A/B on MI355X (profiles/gemm_micro_r02):
+# rocBLAS wins the NT forward and NN dx by 1.5-2x on the tall-skinny
+# bench shapes; OUR split-K MFMA gemm_tn wins the dW reduction by up to
+# 3x (rocBLAS is weak at K >> M=N). "auto" routes each direction to the
+# winner; BNSGCN_GEMM=hip forces the hand-written kernels everywhere.
+_GEMM_MODE = _os.environ.get("BNSGCN_GEMM", "auto")
+
+
 class _Linear(Function):
-    """XW^T + b on the hand-written fp32 MFMA GEMM (K6) when on GPU."""
+    """XW^T + b; see _GEMM_MODE for the MFMA/rocBLAS routing."""
 
     @staticmethod
     def forward(ctx, x, weight, bias):
         ctx.save_for_backward(x, weight)
         ctx.has_bias = bias is not None
-        if use_hip(x):
+        if use_hip(x) and _GEMM_MODE == "hip":
             return get_ext().gemm_nt_bias(x, weight, bias)
         return torch.nn.functional.linear(x, weight, bias)
 
@@ -297,9 +307,10 @@ class _Linear(Function):
         if use_hip(x):
             e = get_ext()
             if ctx.needs_input_grad[0]:
-                gx = e.gemm_nn(grad, weight)
+                gx = e.gemm_nn(grad, weight) if _GEMM_MODE == "hip" \
+                    else grad @ weight
             if ctx.needs_input_grad[1]:
-                gw = e.gemm_tn(grad, x)
+                gw = e.gemm_tn(grad, x)     # split-K: ours wins
             if ctx.has_bias and ctx.needs_input_grad[2]:
                 return gx, gw, e.syncbn_stats(grad)[0]
         else:
