@@ -55,6 +55,7 @@ def parse():
     p.add_argument("--use-pp", action=argparse.BooleanOptionalAction,
                    default=True)
     p.add_argument("--dropout", type=float, default=0.5)
+    p.add_argument("--halo-dtype", choices=["fp32", "bf16"], default="fp32")
     return p.parse_args()
 
 
@@ -87,6 +88,7 @@ def main():
     args.partition_dir = a.partition_dir
     args.use_pp = a.use_pp
     args.dropout = a.dropout
+    args.halo_dtype = a.halo_dtype
     args.eval = False
     args.fix_seed = True
     args.seed = 0

@@ -42,11 +42,16 @@ def comm_stream() -> torch.cuda.Stream:
     return _comm_stream
 
 
-def _exchange(send: torch.Tensor, recv_counts, send_counts) -> torch.Tensor:
+def _exchange(send: torch.Tensor, recv_counts, send_counts,
+              wire_dtype=None) -> torch.Tensor:
+    """wire_dtype=torch.bfloat16 halves the bytes on the xGMI links
+    (--halo-dtype bf16); compute stays fp32 — the cast happens at the
+    wire only (send downcast, recv upcast)."""
+    wire = send if wire_dtype is None else send.to(wire_dtype)
     recv = torch.empty(sum(recv_counts), send.shape[1],
-                       dtype=send.dtype, device=send.device)
-    all_to_all_rows(recv, send, recv_counts, send_counts)
-    return recv
+                       dtype=wire.dtype, device=send.device)
+    all_to_all_rows(recv, wire, recv_counts, send_counts)
+    return recv if wire_dtype is None else recv.float()
 
 
 class _PartitionAggregate(Function):
@@ -72,7 +77,8 @@ class _PartitionAggregate(Function):
                 cs.wait_event(ev)
                 send = pack_rows_raw(x, st.pack_idx, st.pack_scale)
                 with comm_timer.span("forward", cuda=True):
-                    recv = _exchange(send, st.recv_counts, st.send_counts)
+                    recv = _exchange(send, st.recv_counts, st.send_counts,
+                                     plan.wire_dtype)
                 ev_done = cs.record_event()
             out = spmm_sum_raw(inner_indptr, inner_indices, x, src_scale, dst_scale)
             torch.cuda.current_stream().wait_event(ev_done)
@@ -82,7 +88,8 @@ class _PartitionAggregate(Function):
         else:
             send = pack_rows_raw(x, st.pack_idx, st.pack_scale)
             with comm_timer.span("forward"):
-                recv = _exchange(send, st.recv_counts, st.send_counts)
+                recv = _exchange(send, st.recv_counts, st.send_counts,
+                                 plan.wire_dtype)
             out = spmm_sum_raw(inner_indptr, inner_indices, x, src_scale, dst_scale)
             spmm_sum_raw(st.halo_fwd_indptr, st.halo_fwd_indices, recv,
                          src_scale=hscale, dst_scale=dst_scale, out=out)
@@ -106,7 +113,8 @@ class _PartitionAggregate(Function):
                 gr = spmm_sum_raw(st.halo_bwd_indptr, st.halo_bwd_indices, g,
                                   src_scale=dst_scale, dst_scale=hscale)
                 with comm_timer.span("backward", cuda=True):
-                    back = _exchange(gr, st.send_counts, st.recv_counts)
+                    back = _exchange(gr, st.send_counts, st.recv_counts,
+                                     ctx.plan.wire_dtype)
                 ev_done = cs.record_event()
             gx = spmm_sum_raw(inner_t_indptr, inner_t_indices, g,
                               src_scale=dst_scale, dst_scale=src_scale)
@@ -117,7 +125,8 @@ class _PartitionAggregate(Function):
             gr = spmm_sum_raw(st.halo_bwd_indptr, st.halo_bwd_indices, g,
                               src_scale=dst_scale, dst_scale=hscale)
             with comm_timer.span("backward"):
-                back = _exchange(gr, st.send_counts, st.recv_counts)
+                back = _exchange(gr, st.send_counts, st.recv_counts,
+                                 ctx.plan.wire_dtype)
             gx = spmm_sum_raw(inner_t_indptr, inner_t_indices, g,
                               src_scale=dst_scale, dst_scale=src_scale)
             scatter_add_rows_raw(gx, st.pack_idx, back, st.pack_scale)
@@ -141,8 +150,10 @@ class _HaloExchange(Function):
     def forward(ctx, x, plan: HaloPlan, st: EpochState):
         send = pack_rows_raw(x, st.pack_idx, st.pack_scale)
         with comm_timer.span("forward", cuda=x.is_cuda):
-            recv = _exchange(send, st.recv_counts, st.send_counts)
+            recv = _exchange(send, st.recv_counts, st.send_counts,
+                             plan.wire_dtype)
         ctx.st = st
+        ctx.plan = plan
         ctx.n_inner = x.shape[0]
         return recv
 
@@ -151,7 +162,8 @@ class _HaloExchange(Function):
         st: EpochState = ctx.st
         g = g.contiguous()
         with comm_timer.span("backward", cuda=g.is_cuda):
-            back = _exchange(g, st.send_counts, st.recv_counts)
+            back = _exchange(g, st.send_counts, st.recv_counts,
+                             ctx.plan.wire_dtype)
         gx = torch.zeros(ctx.n_inner, g.shape[1], dtype=g.dtype, device=g.device)
         scatter_add_rows_raw(gx, st.pack_idx, back, st.pack_scale)
         return gx, None, None

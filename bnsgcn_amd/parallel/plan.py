@@ -69,7 +69,12 @@ class HaloPlan:
     """One per training job per rank. `set_epoch(e)` refreshes sampling."""
 
     def __init__(self, part: Partition, sampling_rate: float, seed: int,
-                 device: torch.device | str, unit_ratio: bool = False):
+                 device: torch.device | str, unit_ratio: bool = False,
+                 wire_dtype: torch.dtype | None = None):
+        # wire_dtype=torch.bfloat16 (--halo-dtype bf16): halo payloads are
+        # downcast at the wire, halving per-link xGMI bytes; compute and
+        # the 1/ratio estimator stay fp32 (an OPTION — default full fp32)
+        self.wire_dtype = wire_dtype
         self.device = torch.device(device)
         self.rank = part.rank
         self.n_parts = part.n_parts

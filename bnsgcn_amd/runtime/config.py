@@ -75,6 +75,10 @@ def create_parser() -> argparse.ArgumentParser:
                         "rank-0 GPU with the HIP kernels)")
     p.add_argument("--bucket-mb", type=int, default=16,
                    help="gradient all-reduce bucket size (MiB)")
+    p.add_argument("--halo-dtype", "--halo_dtype", choices=["fp32", "bf16"],
+                   default="fp32",
+                   help="wire dtype for the per-layer halo all-to-all: bf16 "
+                        "halves the xGMI bytes; compute stays fp32")
     return p
 
 

@@ -144,9 +144,12 @@ class RankState:
         self.val_mask = torch.from_numpy(part.val_mask).to(dev)
         self.test_mask = torch.from_numpy(part.test_mask).to(dev)
         self.raw_feat = self.feat   # pre-precompute features (dist eval)
+        wire = torch.bfloat16 \
+            if getattr(args, "halo_dtype", "fp32") == "bf16" else None
         self.plan = HaloPlan(part, args.sampling_rate, seed=args.seed, device=dev,
                              unit_ratio=(args.model == "gat"
-                                         and not args.gat_ratio_scale))
+                                         and not args.gat_ratio_scale),
+                             wire_dtype=wire)
         self.ctx = GraphContext.for_partition(part, self.plan, dev,
                                               need_eperm=(args.model == "gat"))
         self.halo_feat0 = None     # GAT use_pp layer-0 full halo features
