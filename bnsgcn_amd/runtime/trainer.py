@@ -436,8 +436,13 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
 
     reducer = GradReducer(model, state.n_train_global,
                           bucket_bytes=args.bucket_mb << 20)
-    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr,
-                                 weight_decay=args.weight_decay)
+    try:        # fused Adam: one multi-tensor kernel instead of ~6 (K11)
+        optimizer = torch.optim.Adam(model.parameters(), lr=args.lr,
+                                     weight_decay=args.weight_decay,
+                                     fused=str(device).startswith("cuda"))
+    except (RuntimeError, ValueError):
+        optimizer = torch.optim.Adam(model.parameters(), lr=args.lr,
+                                     weight_decay=args.weight_decay)
 
     evaluator = None
     pool = None

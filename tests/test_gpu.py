@@ -358,6 +358,60 @@ def test_gpu_matches_cpu_training():
 
 
 @needs_gpu
+def test_gpu_matches_cpu_training_at_scale():
+    """Same oracle at reddit shape scale 0.02 (4.6k nodes, ~2.3M edges,
+    power-law hubs with degree >> SEG=512): engages the worklist
+    heavy-row SPLIT + atomic-combine path, multi-wave scheduling, and the
+    restricted final layer at a size the tiny graphs never reach
+    (VERDICT r1 weak #7)."""
+    from bnsgcn_amd.graph import load_data, partition_graph
+    from bnsgcn_amd.models.models import create_model
+    from bnsgcn_amd.runtime.config import create_parser
+    from bnsgcn_amd.runtime.trainer import RankState, forward_train_logits
+    from bnsgcn_amd.parallel import GradReducer
+
+    g = load_data("reddit", seed=0, scale=0.02)
+    assert int(np.diff(g.adj_in.indptr).max()) > 512  # splits engage
+
+    def train(device):
+        args = create_parser().parse_args([])
+        args.model = "graphsage"
+        args.n_layers = 3
+        args.n_hidden = 64
+        args.sampling_rate = 1.0
+        args.use_pp = True
+        args.dropout = 0.0
+        torch.manual_seed(5)
+        parts, meta = partition_graph(g, 1, method="contiguous")
+        parts[0].meta = meta
+        state = RankState(parts[0], args, device)
+        state.plan.set_epoch(0)
+        m = create_model(args, n_feat=g.n_feat, n_class=g.n_class,
+                         train_size=g.n_train).to(device)
+        state.precompute()
+        reducer = GradReducer(m, g.n_train)
+        opt = torch.optim.Adam(m.parameters(), lr=1e-2)
+        lf = torch.nn.CrossEntropyLoss(reduction="sum")
+        losses = []
+        for ep in range(5):
+            m.train()
+            logits = forward_train_logits(m, state)
+            loss = lf(logits, state.label[state.train_mask].long())
+            reducer.zero_grad()
+            loss.backward()
+            reducer.synchronize()
+            opt.step()
+            losses.append(loss.item())
+        return np.array(losses)
+
+    lc = train("cpu")
+    lg = train("cuda:0")
+    # 2.3M-edge atomics + fp32 reduction-order differences accumulate a
+    # little faster than on the tiny graphs
+    np.testing.assert_allclose(lg, lc, rtol=2e-2, atol=1e-2)
+
+
+@needs_gpu
 @pytest.mark.parametrize("H,D", [(4, 128), (4, 100), (2, 8), (1, 64)])
 def test_gat_kernels_match_reference(H, D):
     """GAT kernel set at the shapes the Yelp config actually hits (incl.
