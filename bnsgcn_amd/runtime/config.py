@@ -83,9 +83,17 @@ def create_parser() -> argparse.ArgumentParser:
 
 
 def graph_name_of(args) -> str:
-    """Reference naming (main.py:18-24)."""
+    """Reference naming (main.py:18-24); a non-unit --data-scale is
+    appended so differently-scaled synthetic stores never collide in the
+    same partition dir (a scale-0.125 run followed by a full-scale run
+    with --skip-partition semantics would silently reuse the small
+    store)."""
     if getattr(args, "graph_name", ""):
         return args.graph_name
     mode = "induc" if args.inductive else "trans"
-    return (f"{args.dataset}-{args.n_partitions}-{args.partition_method}-"
+    name = (f"{args.dataset}-{args.n_partitions}-{args.partition_method}-"
             f"{args.partition_obj}-{mode}")
+    scale = getattr(args, "data_scale", 1.0)
+    if scale != 1.0:
+        name += f"-x{scale:g}"
+    return name
