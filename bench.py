@@ -179,6 +179,9 @@ def main():
     if world > 1:
         dist.barrier()
     elapsed = time.perf_counter() - t0
+    final_loss = step(a.warmup + a.steps)   # untimed: numerics guard
+    assert torch.isfinite(final_loss), \
+        f"non-finite loss after {a.steps} steps — bench result invalid"
 
     # MAX over ranks
     t = torch.tensor([elapsed])
