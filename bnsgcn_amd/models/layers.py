@@ -46,8 +46,22 @@ class LayerNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(dim))
         self.eps = eps
 
+    def forward(self, x, act: bool = False):
+        """act=True fuses the following ReLU into the kernel pair (K9)."""
+        return F.layer_norm(x, self.weight, self.bias, self.eps, act=act)
+
+
+class Dropout(nn.Module):
+    """Drop-in nn.Dropout: mask-free HIP kernel on GPU (K8 — splitmix64
+    mask regenerated in backward), torch on CPU. No parameters, so
+    state-dict keys are unchanged."""
+
+    def __init__(self, p: float):
+        super().__init__()
+        self.p = float(p)
+
     def forward(self, x):
-        return F.layer_norm(x, self.weight, self.bias, self.eps)
+        return F.dropout(x, self.p, self.training)
 
 
 class GCNLayer(nn.Module):
@@ -121,8 +135,8 @@ class GATLayer(nn.Module):
         self.attn_l = nn.Parameter(torch.empty(1, heads, out_feats))
         self.attn_r = nn.Parameter(torch.empty(1, heads, out_feats))
         self.bias = nn.Parameter(torch.zeros(heads * out_feats)) if bias else None
-        self.feat_drop = nn.Dropout(feat_drop)
-        self.attn_drop = nn.Dropout(attn_drop)
+        self.feat_drop = Dropout(feat_drop)
+        self.attn_drop = nn.Dropout(attn_drop)   # p only; fused in softmax2
         self.negative_slope = negative_slope
         gain = math.sqrt(2.0)
         nn.init.xavier_normal_(self.fc.weight, gain=gain)

@@ -12,8 +12,10 @@ from __future__ import annotations
 
 from torch import nn
 
+import torch.nn.functional as TF
+
 from .context import GraphContext
-from .layers import GCNLayer, SAGELayer, GATLayer, LayerNorm
+from .layers import GCNLayer, SAGELayer, GATLayer, LayerNorm, Dropout
 from .sync_bn import SyncBatchNorm
 
 
@@ -35,12 +37,15 @@ class GNNBase(nn.Module):
                                                elementwise_affine=True))
                 else:
                     self.norm.append(SyncBatchNorm(layer_size[i + 1], train_size))
-        self.dropout = nn.Dropout(p=dropout)
+        self.dropout = Dropout(dropout)
 
     def _post(self, i, h):
         if i < self.n_layers - 1:
             if self.use_norm:
-                h = self.norm[i](h)
+                nm = self.norm[i]
+                if isinstance(nm, LayerNorm) and self.activation is TF.relu:
+                    return nm(h, act=True)     # fused LN+ReLU (K7+K9)
+                h = nm(h)
             h = self.activation(h)
         return h
 

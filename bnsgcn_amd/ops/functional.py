@@ -330,26 +330,57 @@ def linear(x, weight, bias=None):
 class _LayerNormF(Function):
     """Row LayerNorm (K7) on the fused gfx950 kernels — torch's ROCm LN
     kernels measured ~27% of the ogbn-products epoch (profiles/
-    topk_products_r02.txt); this pair runs at HBM streaming rate."""
+    topk_products_r02.txt). act=True additionally fuses the inter-layer
+    ReLU (K9): forward emits relu(ln(x)); backward regenerates the
+    activation mask from the saved LN state — no mask, no ReLU kernels."""
 
     @staticmethod
-    def forward(ctx, x, weight, bias, eps):
+    def forward(ctx, x, weight, bias, eps, act):
         x = x.contiguous()
-        y, mean, rstd = get_ext().ln_fwd(x, weight, bias, eps)
-        ctx.save_for_backward(x, weight, mean, rstd)
+        y, mean, rstd = get_ext().ln_fwd(x, weight, bias, eps, int(act))
+        ctx.save_for_backward(x, weight, bias, mean, rstd)
+        ctx.act = int(act)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w, mean, rstd = ctx.saved_tensors
-        dx, dw, db = get_ext().ln_bwd(x, dy.contiguous(), w, mean, rstd)
-        return dx, dw, db, None
+        x, w, b, mean, rstd = ctx.saved_tensors
+        dx, dw, db = get_ext().ln_bwd(x, dy.contiguous(), w, b, mean, rstd,
+                                      ctx.act)
+        return dx, dw, db, None, None
 
 
-def layer_norm(x, weight, bias, eps: float = 1e-5):
+def layer_norm(x, weight, bias, eps: float = 1e-5, act: bool = False):
+    """act=True computes relu(layer_norm(x)) in one fused pair on GPU."""
     if use_hip(x) and x.shape[-1] <= 1024 and x.dtype == torch.float32:
-        return _LayerNormF.apply(x, weight, bias, eps)
-    return torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+        return _LayerNormF.apply(x, weight, bias, eps, act)
+    y = torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+    return torch.relu(y) if act else y
+
+
+class _DropoutF(Function):
+    """Mask-free dropout (K8): the splitmix64 mask is regenerated from the
+    saved seed in backward — one elementwise kernel each way, no mask
+    tensor, no torch RNG-state kernels."""
+
+    @staticmethod
+    def forward(ctx, x, keep, seed):
+        ctx.keep, ctx.seed = keep, seed
+        return get_ext().dropout_apply(x.contiguous(), keep, seed)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return (get_ext().dropout_apply(dy.contiguous(), ctx.keep, ctx.seed),
+                None, None)
+
+
+def dropout(x, p: float, training: bool):
+    if not training or p <= 0.0:
+        return x
+    if use_hip(x) and x.dtype == torch.float32:
+        seed = int(torch.randint(0, 2**62, (1,)).item())
+        return _DropoutF.apply(x, 1.0 - p, seed)
+    return torch.nn.functional.dropout(x, p, training)
 
 
 def segment_softmax2_raw(indptr1, logits1, indptr2, logits2):

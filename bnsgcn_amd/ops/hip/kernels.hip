@@ -989,11 +989,15 @@ __global__ void attn_project_dattn_kernel(
 
 #define LN_MAX_K 16  // supports F up to 64*16
 
+// act=1 fuses the inter-layer ReLU (model _post pairs them): forward
+// writes max(ln, 0); backward recomputes the pre-activation from
+// (x, mean, rstd, gamma, beta) and masks dy — no mask storage, no
+// separate ReLU kernels (K9).
 __global__ __launch_bounds__(256) void ln_fwd_kernel(
     const float* __restrict__ x, const float* __restrict__ gamma,
     const float* __restrict__ beta, float* __restrict__ y,
     float* __restrict__ mean_out, float* __restrict__ rstd_out,
-    int64_t n, int F, float eps) {
+    int64_t n, int F, float eps, int act) {
   const int lane = threadIdx.x & 63;
   const float inv_f = 1.0f / F;
   float vals[LN_MAX_K], gm[LN_MAX_K], bt[LN_MAX_K];
@@ -1014,23 +1018,29 @@ __global__ __launch_bounds__(256) void ln_fwd_kernel(
     const float rstd = rsqrtf(v * inv_f + eps);
     float* yr = y + r * (int64_t)F;
     k = 0;
-    for (int f = lane; f < F; f += WAVE, ++k)
-      yr[f] = (vals[k] - mu) * rstd * gm[k] + bt[k];
+    for (int f = lane; f < F; f += WAVE, ++k) {
+      const float o = (vals[k] - mu) * rstd * gm[k] + bt[k];
+      yr[f] = act ? fmaxf(o, 0.f) : o;
+    }
     if (lane == 0) { mean_out[r] = mu; rstd_out[r] = rstd; }
   }
 }
 
 // dx = rstd * ( g - mean(g) - xhat * mean(g * xhat) ),  g = dy * gamma
+// (act=1: g is first masked by the recomputed pre-activation sign)
 __global__ __launch_bounds__(256) void ln_bwd_dx_kernel(
     const float* __restrict__ x, const float* __restrict__ dy,
-    const float* __restrict__ gamma, const float* __restrict__ mean,
-    const float* __restrict__ rstd, float* __restrict__ dx,
-    int64_t n, int F) {
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    float* __restrict__ dx, int64_t n, int F, int act) {
   const int lane = threadIdx.x & 63;
   const float inv_f = 1.0f / F;
-  float xh[LN_MAX_K], g[LN_MAX_K], gm[LN_MAX_K];
+  float xh[LN_MAX_K], g[LN_MAX_K], gm[LN_MAX_K], bt[LN_MAX_K];
   int K = 0;
-  for (int f = lane; f < F; f += WAVE, ++K) gm[K] = gamma[f];
+  for (int f = lane; f < F; f += WAVE, ++K) {
+    gm[K] = gamma[f];
+    bt[K] = act ? beta[f] : 0.f;
+  }
   const int64_t stride = (int64_t)gridDim.x * 4;
   for (int64_t r = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6); r < n;
        r += stride) {
@@ -1041,7 +1051,9 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_kernel(
     int k = 0;
     for (int f = lane; f < F; f += WAVE, ++k) {
       xh[k] = (xr[f] - mu) * rs;
-      g[k] = dr[f] * gm[k];
+      float d = dr[f];
+      if (act && xh[k] * gm[k] + bt[k] <= 0.f) d = 0.f;
+      g[k] = d * gm[k];
       a += g[k];
       b += g[k] * xh[k];
     }
@@ -1055,24 +1067,45 @@ __global__ __launch_bounds__(256) void ln_bwd_dx_kernel(
 }
 
 // dgamma[f] = sum_r dy*xhat, dbeta[f] = sum_r dy — column reduction in
-// the syncbn_stats shape (thread per column, chunked rows, atomic combine)
+// the syncbn_stats shape (thread per column, chunked rows, atomic
+// combine); act=1 masks dy by the recomputed pre-activation sign
 __global__ void ln_bwd_w_kernel(
     const float* __restrict__ x, const float* __restrict__ dy,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ out, int64_t n, int F, int row_chunks) {
+    float* __restrict__ out, int64_t n, int F, int row_chunks, int act) {
   const int f = blockIdx.x * blockDim.x + threadIdx.x;
   if (f >= F) return;
+  const float gmf = act ? gamma[f] : 0.f;
+  const float btf = act ? beta[f] : 0.f;
   const int64_t rows_per = (n + row_chunks - 1) / row_chunks;
   const int64_t r0 = (int64_t)blockIdx.y * rows_per;
   const int64_t r1 = (r0 + rows_per < n) ? r0 + rows_per : n;
   float sg = 0.f, sb = 0.f;
   for (int64_t r = r0; r < r1; ++r) {
-    const float d = dy[r * (int64_t)F + f];
-    sg += d * (x[r * (int64_t)F + f] - mean[r]) * rstd[r];
+    const float xh = (x[r * (int64_t)F + f] - mean[r]) * rstd[r];
+    float d = dy[r * (int64_t)F + f];
+    if (act && xh * gmf + btf <= 0.f) d = 0.f;
+    sg += d * xh;
     sb += d;
   }
   if (row_chunks == 1) { out[f] = sg; out[F + f] = sb; }
   else { atomicAdd(&out[f], sg); atomicAdd(&out[F + f], sb); }
+}
+
+// -------------------------- dropout (K8) -------------------------------
+// Mask-free feature dropout: out[i] = keep(seed, i) ? x[i]/keep : 0,
+// mask regenerated from splitmix64 — the SAME kernel serves forward and
+// backward (apply to x, then to dy), so no mask tensor ever exists.
+__global__ __launch_bounds__(256) void dropout_apply_kernel(
+    const float* __restrict__ x, float* __restrict__ out, int64_t n,
+    float keep, uint64_t seed) {
+  const uint32_t thr = (uint32_t)(keep * 4294967296.0);
+  const float inv_keep = 1.0f / keep;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = drop_keep(i, thr, seed) ? x[i] * inv_keep : 0.f;
 }
 
 // ------------------------------ launchers ------------------------------
@@ -1219,7 +1252,7 @@ at::Tensor syncbn_stats(at::Tensor x) {
 }
 
 std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor gamma,
-                               at::Tensor beta, double eps) {
+                               at::Tensor beta, double eps, int64_t act) {
   check_f32(x, "x");
   check_f32(gamma, "gamma");
   check_f32(beta, "beta");
@@ -1235,12 +1268,13 @@ std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor gamma,
                      stream, x.data_ptr<float>(), gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), y.data_ptr<float>(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(), n, F,
-                     (float)eps);
+                     (float)eps, (int)act);
   return {y, mean, rstd};
 }
 
 std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
-                               at::Tensor mean, at::Tensor rstd) {
+                               at::Tensor beta, at::Tensor mean,
+                               at::Tensor rstd, int64_t act) {
   check_f32(x, "x");
   check_f32(dy, "dy");
   const int64_t n = x.size(0);
@@ -1251,15 +1285,30 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor gamma,
   const int blocks = (int)std::min<int64_t>((n + 3) / 4, 32768);
   hipLaunchKernelGGL(ln_bwd_dx_kernel, dim3(std::max(blocks, 1)), dim3(256),
                      0, stream, x.data_ptr<float>(), dy.data_ptr<float>(),
-                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
-                     rstd.data_ptr<float>(), dx.data_ptr<float>(), n, F);
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     dx.data_ptr<float>(), n, F, (int)act);
   const int row_chunks = (int)std::min<int64_t>((n + 255) / 256, 2048);
   dim3 grid((F + 255) / 256, row_chunks);
   hipLaunchKernelGGL(ln_bwd_w_kernel, grid, dim3(256), 0, stream,
                      x.data_ptr<float>(), dy.data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     dw.data_ptr<float>(), n, F, row_chunks);
+                     dw.data_ptr<float>(), n, F, row_chunks, (int)act);
   return {dx, dw[0], dw[1]};
+}
+
+at::Tensor dropout_apply(at::Tensor x, double keep, int64_t seed) {
+  check_f32(x, "x");
+  auto out = at::empty_like(x);
+  const int64_t n = x.numel();
+  if (n == 0) return out;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int blocks = (int)std::min<int64_t>((n + 255) / 256, 32768);
+  hipLaunchKernelGGL(dropout_apply_kernel, dim3(blocks), dim3(256), 0,
+                     stream, x.data_ptr<float>(), out.data_ptr<float>(), n,
+                     (float)keep, (uint64_t)seed);
+  return out;
 }
 
 at::Tensor bincount_i32(at::Tensor v, int64_t n_bins) {
@@ -1581,8 +1630,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("spmm_edge_sum", &spmm_edge_sum, "multi-head edge-weighted SpMM");
   m.def("sddmm_dot", &sddmm_dot, "per-edge per-head dot (spmm_edge grad)");
   m.def("bincount_i32", &bincount_i32, "atomic int32 histogram");
-  m.def("ln_fwd", &ln_fwd, "row LayerNorm forward (y, mean, rstd)");
+  m.def("ln_fwd", &ln_fwd, "row LayerNorm forward (y, mean, rstd); "
+        "act=1 fuses ReLU");
   m.def("ln_bwd", &ln_bwd, "row LayerNorm backward (dx, dgamma, dbeta)");
+  m.def("dropout_apply", &dropout_apply,
+        "mask-free dropout (splitmix64; same kernel fwd and bwd)");
   m.def("attn_project", &attn_project, "fused GAT el/er projections");
   m.def("attn_project_backward", &attn_project_backward,
         "fused GAT projection backward");

@@ -232,16 +232,18 @@ def test_fused_attn_dropout_softmax2():
 
 @needs_gpu
 @pytest.mark.parametrize("F", [41, 64, 128, 256, 600, 1024])
-def test_layernorm_matches_torch(F):
-    """HIP ln_fwd/ln_bwd (K7) vs torch LayerNorm: outputs and all three
-    gradients, including F not a multiple of the 64-lane wave."""
+@pytest.mark.parametrize("act", [False, True])
+def test_layernorm_matches_torch(F, act):
+    """HIP ln_fwd/ln_bwd (K7, + fused ReLU when act — K9) vs torch
+    LayerNorm(+relu): outputs and all three gradients, including F not a
+    multiple of the 64-lane wave."""
     from bnsgcn_amd.ops.functional import layer_norm
     torch.manual_seed(0)
     n = 4097
     x = (torch.randn(n, F, device="cuda:0") * 3 + 1).requires_grad_(True)
     w = torch.randn(F, device="cuda:0").requires_grad_(True)
     b = torch.randn(F, device="cuda:0").requires_grad_(True)
-    y = layer_norm(x, w, b)
+    y = layer_norm(x, w, b, act=act)
     g = torch.randn_like(y)
     y.backward(g)
 
@@ -249,11 +251,35 @@ def test_layernorm_matches_torch(F):
     w2 = w.detach().clone().requires_grad_(True)
     b2 = b.detach().clone().requires_grad_(True)
     y2 = torch.nn.functional.layer_norm(x2, (F,), w2, b2, 1e-5)
+    if act:
+        y2 = torch.relu(y2)
     y2.backward(g)
     torch.testing.assert_close(y, y2, rtol=2e-5, atol=2e-5)
     torch.testing.assert_close(x.grad, x2.grad, rtol=2e-4, atol=2e-4)
     torch.testing.assert_close(w.grad, w2.grad, rtol=2e-3, atol=2e-3)
     torch.testing.assert_close(b.grad, b2.grad, rtol=2e-3, atol=2e-3)
+
+
+@needs_gpu
+def test_hip_dropout_unbiased_and_consistent():
+    """Mask-free HIP dropout (K8): kept fraction ~= keep, kept values
+    scaled 1/keep, and backward applies the SAME regenerated mask."""
+    from bnsgcn_amd.ops.functional import dropout
+    torch.manual_seed(0)
+    x = torch.rand(1000, 257, device="cuda:0") + 1.0   # strictly positive
+    x.requires_grad_(True)
+    p = 0.4
+    y = dropout(x, p, training=True)
+    mask = (y != 0).float()
+    frac = float(mask.mean())
+    assert abs(frac - (1 - p)) < 0.01, frac
+    torch.testing.assert_close(y, x * mask / (1 - p), rtol=1e-6, atol=1e-7)
+    g = torch.randn_like(y)
+    y.backward(g)
+    torch.testing.assert_close(x.grad, g * mask / (1 - p),
+                               rtol=1e-6, atol=1e-7)
+    # eval mode: identity
+    assert dropout(x, p, training=False) is x
 
 
 @needs_gpu
