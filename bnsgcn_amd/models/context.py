@@ -235,6 +235,43 @@ class GraphContext:
         return (st.halo_fwd_indptr, st.halo_fwd_indices,
                 st.halo_bwd_indptr, st.halo_bwd_indices, st.halo_eperm_t)
 
+    def gat_rows_inner(self):
+        """Final-layer loss-row restriction for GAT: row-gathered INNER
+        edge set (ip, ix, tip, tix, eperm_t) — static, cached."""
+        if getattr(self, "_gat_rows_inner", None) is None:
+            from ..ops.csr_torch import gather_rows_csr
+            rows = self.loss_rows
+            ip, ix = gather_rows_csr(self.indptr, self.indices, rows)
+            # transpose_csr's eperm maps t-edge -> fwd-edge, which is
+            # exactly the wperm the backward spmm_edge consumes
+            tip, tix, eperm_t = transpose_csr(ip, ix, self.plan.n_inner)
+            if ip.is_cuda:
+                from ..ops.functional import _worklist_of
+                _worklist_of(ip)
+                _worklist_of(tip)
+            self._gat_rows_inner = (ip, ix, tip, tix, eperm_t)
+        return self._gat_rows_inner
+
+    def gat_rows_halo(self, st: EpochState):
+        """Row-gathered sampled-halo edge set for the restricted final
+        GAT layer, cached per epoch."""
+        c = getattr(self, "_gat_rows_halo", None)
+        if c is not None and c[0] == st.epoch:
+            return c[1]
+        from ..ops.csr_torch import gather_rows_csr
+        rows = self.loss_rows
+        hip_, hix_ = gather_rows_csr(st.halo_fwd_indptr,
+                                     st.halo_fwd_indices, rows)
+        R = int(st.halo_bwd_indptr.numel() - 1)
+        hbip, hbix, eperm_t = transpose_csr(hip_, hix_, R)
+        if hip_.is_cuda:
+            from ..ops.functional import _worklist_of
+            _worklist_of(hip_)
+            _worklist_of(hbip)
+        out = (hip_, hix_, hbip, hbix, eperm_t)
+        self._gat_rows_halo = (st.epoch, out)
+        return out
+
     def gat_split_full(self):
         """Static FULL-halo edge set (GAT layer 0 under use_pp)."""
         if getattr(self, "_gat_split_full_cache", None) is None:

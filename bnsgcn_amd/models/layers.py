@@ -143,7 +143,8 @@ class GATLayer(nn.Module):
         nn.init.xavier_normal_(self.attn_l, gain=gain)
         nn.init.xavier_normal_(self.attn_r, gain=gain)
 
-    def forward(self, ctx: GraphContext, x, halo_feat: torch.Tensor | None = None):
+    def forward(self, ctx: GraphContext, x, halo_feat: torch.Tensor | None = None,
+                rows: torch.Tensor | None = None):
         """x: [n_local, F]. In partition mode, halo sources are fetched via
         halo_exchange — except when `halo_feat` is given (GAT layer 0 under
         use_pp: the FULL unsampled halo features captured at precompute,
@@ -181,18 +182,25 @@ class GATLayer(nn.Module):
         el_in, er = F.attn_project(z_in, self.attn_l, self.attn_r)
         el_h, _ = F.attn_project(z_h, self.attn_l, self.attn_r)  # halo: el
         slope = self.negative_slope
-        li = F.sddmm_add(el_in, er, ctx.indptr, ctx.indices,
-                         ctx.t_indptr, ctx.t_indices, ctx.t_eperm, slope=slope)
+        if rows is not None:
+            # final-layer loss-row restriction: edge sets gathered to the
+            # labeled dst rows (identical math — unlabeled rows carry zero
+            # gradient); sources (z/el) stay full, er shrinks to the rows
+            iip, iix, itip, itix, ieperm = ctx.gat_rows_inner()
+            hip, hix, hbip, hbix, heperm = ctx.gat_rows_halo(ctx.plan.state)
+            er = er[rows]
+            inner5 = (iip, iix, itip, itix, ieperm)
+        else:
+            inner5 = (ctx.indptr, ctx.indices, ctx.t_indptr, ctx.t_indices,
+                      ctx.t_eperm)
+        li = F.sddmm_add(el_in, er, *inner5, slope=slope)
         lh = F.sddmm_add(el_h, er, hip, hix, hbip, hbix, heperm, slope=slope)
         # attention dropout FUSED into the union softmax (no separate
         # dropout kernels/masks; identity when eval or p=0)
         p_attn = self.attn_drop.p if self.training else 0.0
-        ai, ah = F.segment_softmax2(li, lh, ctx.indptr, hip, p_drop=p_attn)
-        out = F.spmm_edge_sum2(
-            z_in, ai, z_h, ah,
-            (ctx.indptr, ctx.indices, ctx.t_indptr, ctx.t_indices,
-             ctx.t_eperm),
-            (hip, hix, hbip, hbix, heperm))
+        ai, ah = F.segment_softmax2(li, lh, inner5[0], hip, p_drop=p_attn)
+        out = F.spmm_edge_sum2(z_in, ai, z_h, ah, inner5,
+                               (hip, hix, hbip, hbix, heperm))
         if self.bias is not None:
             out = out + self.bias.view(1, H, D)
         return out

@@ -108,21 +108,32 @@ class GAT(GNNBase):
 
     def forward(self, ctx: GraphContext, feat, halo_feat0=None):
         """halo_feat0: full-halo raw features for layer 0 under use_pp
-        (captured once at precompute; reference train.py:208-209)."""
+        (captured once at precompute; reference train.py:208-209).
+        Training with ctx.loss_rows set restricts the FINAL layer to the
+        labeled rows (same math; see GNNBase/GCN.forward)."""
         h = feat
+        restrict = (self.training and ctx.plan is not None
+                    and getattr(ctx, "loss_rows", None) is not None)
         for i in range(self.n_layers):
+            last_conv_restrict = (restrict and i == self.n_layers - 1
+                                  and not (self.use_pp and i == 0))
             if i < self.n_layers - self.n_linear:
                 lay = self.layers[i]
+                rows = ctx.loss_rows if last_conv_restrict else None
                 if (self.training and self.use_pp and i == 0
                         and ctx.plan is not None):
                     h = lay(ctx, h, halo_feat=halo_feat0)
                 else:
-                    h = lay(ctx, h)
+                    h = lay(ctx, h, rows=rows)
                 h = h.mean(1)  # head mean (model.py:124)
             else:
                 h = self.dropout(h)
+                if restrict and i == self.n_layers - 1:
+                    h = h[ctx.loss_rows]
                 h = self.layers[i](h)
             h = self._post(i, h)
+        if restrict and h.shape[0] != ctx.loss_rows.shape[0]:
+            h = h[ctx.loss_rows]   # 1-layer use_pp edge case
         return h
 
 

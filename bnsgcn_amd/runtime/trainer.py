@@ -155,13 +155,13 @@ class RankState:
         self.halo_feat0 = None     # GAT use_pp layer-0 full halo features
         self.n_train_global = int(part.meta["n_train"])
         self.part_train = int(part.train_mask.sum())
-        # final-layer loss-row restriction (GCN/SAGE): train-time logits are
-        # computed only for labeled rows — identical loss/gradients, and
-        # [N, C] logits shrink to [n_train_local, C] (papers100M: 76 GB ->
-        # 0.8 GB). BNSGCN_FULL_LOGITS=1 restores the reference's full pass.
+        # final-layer loss-row restriction (all models): train-time logits
+        # are computed only for labeled rows — identical loss/gradients,
+        # and [N, C] logits shrink to [n_train_local, C] (papers100M:
+        # 76 GB -> 0.8 GB). BNSGCN_FULL_LOGITS=1 restores the reference's
+        # full pass.
         self.loss_rows = None
-        if (os.environ.get("BNSGCN_FULL_LOGITS") != "1"
-                and args.model != "gat"):
+        if os.environ.get("BNSGCN_FULL_LOGITS") != "1":
             self.loss_rows = torch.nonzero(self.train_mask).flatten()
             self.ctx.loss_rows = self.loss_rows
 
