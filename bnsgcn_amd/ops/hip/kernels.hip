@@ -1151,8 +1151,12 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
     return e ? atoi(e) : 0;  // measured: strided LOSES ~1-6%
     // (drifting waves break their own temporal locality; bench_strided_*)
   }();
+  static const int subw32_max = [] {
+    const char* e = getenv("BNSGCN_SPMM_SUBW32_MAX");
+    return e ? atoi(e) : 32;  // f4 <= this runs the half-wave variant
+  }();
   if (F % 4 == 0) {
-    if (F / 4 <= 32 && n_waves % 8 == 0) {
+    if (F / 4 <= subw32_max && n_waves % 8 == 0) {
       const int strided = strided_env && (n_waves % 8 == 0);
       auto kfn = acc ? spmm_sum_vec4_kernel<true, 32>
                      : spmm_sum_vec4_kernel<false, 32>;
