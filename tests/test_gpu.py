@@ -283,6 +283,66 @@ def test_hip_dropout_unbiased_and_consistent():
 
 
 @needs_gpu
+def test_stream_overlap_ordering_stress(monkeypatch):
+    """Stream-ordering stress: 25 sampled epochs with the
+    side-stream halo overlap ON must be BITWISE identical to the
+    sequential path (BNSGCN_NO_OVERLAP=1) and run-to-run deterministic.
+    Heavy-row splits are disabled (SEG=1e9 -> no atomicAdd combines) so
+    every kernel is deterministic — any missing cross-stream event
+    (exchange vs SpMM, prefetch vs consume) shows up as a value change."""
+    from bnsgcn_amd.graph import load_data, partition_graph
+    from bnsgcn_amd.models.models import create_model
+    from bnsgcn_amd.runtime.config import create_parser
+    from bnsgcn_amd.runtime.trainer import RankState, forward_train_logits
+    from bnsgcn_amd.parallel import GradReducer
+    from bnsgcn_amd.ops import csr_torch
+    monkeypatch.setattr(csr_torch, "SEG", 10**9)
+
+    def train(no_overlap: bool):
+        monkeypatch.setenv("BNSGCN_NO_OVERLAP", "1" if no_overlap else "0")
+        args = create_parser().parse_args([])
+        args.dataset = "tiny"
+        args.model = "gcn"
+        args.n_layers = 3
+        args.n_hidden = 32
+        args.sampling_rate = 0.4
+        args.use_pp = False          # layer 0 exchanges too (more traffic)
+        args.dropout = 0.2
+        torch.manual_seed(11)
+        g = load_data("tiny", seed=3)
+        parts, meta = partition_graph(g, 1, method="metis")
+        parts[0].meta = meta
+        state = RankState(parts[0], args, "cuda:0")
+        state.plan.set_epoch(0)
+        m = create_model(args, n_feat=g.n_feat, n_class=g.n_class,
+                         train_size=g.n_train).to("cuda:0")
+        reducer = GradReducer(m, g.n_train)
+        opt = torch.optim.Adam(m.parameters(), lr=1e-2, fused=True)
+        lf = torch.nn.CrossEntropyLoss(reduction="sum")
+        losses = []
+        for ep in range(25):
+            state.plan.set_epoch(ep)
+            m.train()
+            logits = forward_train_logits(m, state)
+            loss = lf(logits, state.label[state.train_mask].long())
+            reducer.zero_grad()
+            loss.backward()
+            if not no_overlap:
+                state.prefetch(ep + 1)
+            reducer.synchronize()
+            opt.step()
+            losses.append(loss.item())
+        torch.cuda.synchronize()
+        return np.array(losses)
+
+    a1 = train(False)
+    a2 = train(False)
+    b = train(True)
+    np.testing.assert_array_equal(a1, a2)   # run-to-run deterministic
+    np.testing.assert_array_equal(a1, b)    # overlap == sequential
+
+
+@needs_gpu
 def test_plan_prefetch_trajectory_identical():
     """Side-stream plan prefetch (RankState.prefetch) must be trajectory-
     identical to building the sampling plan on the main stream — same
