@@ -99,6 +99,10 @@ def build_worklist(indptr: torch.Tensor, seg: int = None, max_waves: int = None
     lens = (wend - wbeg)
     cum = torch.cumsum(torch.cat([torch.zeros(1, dtype=lens.dtype, device=device),
                                   lens]), 0)
+    # graphs with millions of items want more waves in flight: 262144
+    # beat 131072 by ~3% on ogbn-products (2.7M items; profiles r02)
+    if total >= 1_000_000:
+        max_waves = max(max_waves, 262144)
     n_waves = min(max_waves, max(8, total))
     n_waves = (n_waves + 7) // 8 * 8
     targets = (cum[-1] * torch.arange(n_waves + 1, device=device).double()
