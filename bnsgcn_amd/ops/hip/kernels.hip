@@ -705,11 +705,16 @@ __global__ __launch_bounds__(256) void spmm_edge_vec4_kernel(
     const bool atomic = row < 0;
     if (atomic) row = ~row;
     const int64_t beg = wbeg[it], end = wend[it];
-    for (int f0 = 0; f0 < HD4; f0 += WAVE) {
+    // two float4 accumulators per lane (like spmm_sum_vec4): HD4 <= 128
+    // — the common [H=4, D=128] GAT shape — walks the edge list ONCE
+    // instead of once per 64-float4 feature pass
+    for (int f0 = 0; f0 < HD4; f0 += 2 * WAVE) {
       const int fA = f0 + lane;
-      const bool hasA = fA < HD4;
+      const int fB = fA + WAVE;
+      const bool hasA = fA < HD4, hasB = fB < HD4;
       const int hA = hasA ? fA / D4 : 0;
-      float4 acc = {0.f, 0.f, 0.f, 0.f};
+      const int hB = hasB ? fB / D4 : 0;
+      float4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
       for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
         const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
         int cid = 0;
@@ -717,25 +722,38 @@ __global__ __launch_bounds__(256) void spmm_edge_vec4_kernel(
 #pragma unroll 4
         for (int k = 0; k < nv; ++k) {
           const int c = __shfl(cid, k, WAVE);
-          if (hasA) {
-            const int64_t we = wperm ? wperm[e0 + k] : (e0 + k);
-            const float ww = w[we * H + hA];
-            f4_axpy(acc, ww, x4[(int64_t)c * HD4 + fA]);
-          }
+          const int64_t we = wperm ? wperm[e0 + k] : (e0 + k);
+          const int64_t base = (int64_t)c * HD4;
+          if (hasA) f4_axpy(acc0, w[we * H + hA], x4[base + fA]);
+          if (hasB) f4_axpy(acc1, w[we * H + hB], x4[base + fB]);
         }
       }
-      if (!hasA) continue;
-      const int64_t o = (int64_t)row * HD4 + fA;
+      const int64_t ob = (int64_t)row * HD4;
       if (atomic) {
-        float* p = reinterpret_cast<float*>(&out4[o]);
-        atomicAdd(p + 0, acc.x); atomicAdd(p + 1, acc.y);
-        atomicAdd(p + 2, acc.z); atomicAdd(p + 3, acc.w);
+        if (hasA) {
+          float* p = reinterpret_cast<float*>(&out4[ob + fA]);
+          atomicAdd(p + 0, acc0.x); atomicAdd(p + 1, acc0.y);
+          atomicAdd(p + 2, acc0.z); atomicAdd(p + 3, acc0.w);
+        }
+        if (hasB) {
+          float* p = reinterpret_cast<float*>(&out4[ob + fB]);
+          atomicAdd(p + 0, acc1.x); atomicAdd(p + 1, acc1.y);
+          atomicAdd(p + 2, acc1.z); atomicAdd(p + 3, acc1.w);
+        }
       } else if (ACC) {
-        float4 pv = out4[o];
-        pv.x += acc.x; pv.y += acc.y; pv.z += acc.z; pv.w += acc.w;
-        out4[o] = pv;
+        if (hasA) {
+          float4 pv = out4[ob + fA];
+          pv.x += acc0.x; pv.y += acc0.y; pv.z += acc0.z; pv.w += acc0.w;
+          out4[ob + fA] = pv;
+        }
+        if (hasB) {
+          float4 pv = out4[ob + fB];
+          pv.x += acc1.x; pv.y += acc1.y; pv.z += acc1.z; pv.w += acc1.w;
+          out4[ob + fB] = pv;
+        }
       } else {
-        out4[o] = acc;
+        if (hasA) out4[ob + fA] = acc0;
+        if (hasB) out4[ob + fB] = acc1;
       }
     }
   }
