@@ -614,9 +614,9 @@ def test_train_and_eval_on_gpu_accuracy():
 
 
 @needs_gpu
-def test_segment_softmax2_matches_reference():
+@pytest.mark.parametrize("H", [4, 3, 1, 16])   # pow2 -> interleaved kernel
+def test_segment_softmax2_matches_reference(H):
     from bnsgcn_amd.ops.functional import segment_softmax2_raw
-    H = 4
     ip1, _ = rand_csr(300, 10, 4000, seed=41)
     ip2, _ = rand_csr(300, 10, 700, seed=42)
     l1 = torch.randn(4000, H)
