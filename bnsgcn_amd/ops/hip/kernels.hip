@@ -1508,7 +1508,15 @@ std::vector<at::Tensor> segment_softmax2(at::Tensor ip1, at::Tensor l1,
   auto da2 = drop ? at::empty_like(l2) : a2;
   if (n_rows == 0) return {a1, a2, da1, da2};
   auto stream = at::cuda::getCurrentCUDAStream();
-  const bool pow2 = H <= 32 && (H & (H - 1)) == 0;
+  // interleaved variant: coalesced but consolidates per-(row,head) waves
+  // into per-row ones — measured SLOWER on Yelp GAT (85 -> 112 ms epoch;
+  // short power-law segments are latency-bound and lose the 4x head
+  // parallelism). Kept behind BNSGCN_SOFTMAX_ILV=1.
+  static const int ilv_env = [] {
+    const char* e = getenv("BNSGCN_SOFTMAX_ILV");
+    return e ? atoi(e) : 0;
+  }();
+  const bool pow2 = ilv_env && H <= 32 && (H & (H - 1)) == 0;
   if (pow2) {
     hipLaunchKernelGGL(segment_softmax2_ilv_kernel,
                        dim3(spmm_grid(n_rows)), dim3(256), 0, stream,
@@ -1539,7 +1547,11 @@ std::vector<at::Tensor> segment_softmax2_backward(at::Tensor ip1, at::Tensor a1,
   auto d2 = at::empty_like(a2);
   if (n_rows == 0) return {d1, d2};
   auto stream = at::cuda::getCurrentCUDAStream();
-  const bool pow2 = H <= 32 && (H & (H - 1)) == 0;
+  static const int ilv_env = [] {
+    const char* e = getenv("BNSGCN_SOFTMAX_ILV");
+    return e ? atoi(e) : 0;
+  }();
+  const bool pow2 = ilv_env && H <= 32 && (H & (H - 1)) == 0;
   if (pow2) {
     hipLaunchKernelGGL(segment_softmax2_ilv_bwd_kernel,
                        dim3(spmm_grid(n_rows)), dim3(256), 0, stream,
