@@ -270,8 +270,11 @@ def test_fuzz_configs(tmp_path, seed):
             n_epochs=3,
             dropout=rng.choice([0.0, 0.3]),
             inductive=rng.choice([True, False]),
-            partition_method=rng.choice(["metis", "random"]),
+            partition_method=rng.choice(["metis", "random", "bfs",
+                                         "contiguous"]),
             n_linear=rng.choice([0, 1]),
+            halo_dtype=rng.choice(["fp32", "bf16"]),
+            norm=rng.choice(["layer", "layer", "none"]),
         )
         res = _run_config(tmp_path / f"f{seed}_{trial}", world, **kw)
         for m in res:
