@@ -511,6 +511,179 @@ DEV_INLINE bool drop_keep(int64_t idx, uint32_t thr, uint64_t seed) {
   return (uint32_t)(z >> 32) < thr;
 }
 
+DEV_INLINE float4 f4_max(float4 a, float4 b) {
+  return make_float4(fmaxf(a.x, b.x), fmaxf(a.y, b.y), fmaxf(a.z, b.z),
+                     fmaxf(a.w, b.w));
+}
+
+DEV_INLINE float4 wave_reduce_sum4(float4 v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    v.x += __shfl_xor(v.x, off, WAVE);
+    v.y += __shfl_xor(v.y, off, WAVE);
+    v.z += __shfl_xor(v.z, off, WAVE);
+    v.w += __shfl_xor(v.w, off, WAVE);
+  }
+  return v;
+}
+
+DEV_INLINE float4 wave_reduce_max4(float4 v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float4 o = make_float4(__shfl_xor(v.x, off, WAVE),
+                           __shfl_xor(v.y, off, WAVE),
+                           __shfl_xor(v.z, off, WAVE),
+                           __shfl_xor(v.w, off, WAVE));
+    v = f4_max(v, o);
+  }
+  return v;
+}
+
+// H == 4 (the GAT bench head count): one wave per row, each LANE loads
+// one edge's float4 of ALL FOUR heads — fully coalesced [E,4] access
+// with the SAME per-row pass count as the per-(row,head) form (64 edges
+// per pass; the head work rides in the lane's float4 ALU). The
+// per-(row,head) form reads at stride 16 B (1/4 cacheline utilization)
+// and measured ~20% of the Yelp GAT epoch.
+__global__ void segment_softmax2_h4_kernel(
+    const int64_t* __restrict__ ip1, const float4* __restrict__ l1,
+    const int64_t* __restrict__ ip2, const float4* __restrict__ l2,
+    float4* __restrict__ a1, float4* __restrict__ a2,
+    float4* __restrict__ da1, float4* __restrict__ da2,
+    int n_rows, float keep, uint64_t seed, int64_t off2) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const bool drop = keep < 1.0f;
+  const uint32_t thr = (uint32_t)(keep * 4294967296.0);
+  const float inv_keep = drop ? 1.0f / keep : 1.0f;
+  for (int r = wave; r < n_rows; r += n_waves) {
+    const int64_t b1 = ip1[r], e1 = ip1[r + 1];
+    const int64_t b2 = ip2[r], e2 = ip2[r + 1];
+    if (b1 == e1 && b2 == e2) continue;
+    float4 m = make_float4(-INFINITY, -INFINITY, -INFINITY, -INFINITY);
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) m = f4_max(m, l1[e]);
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) m = f4_max(m, l2[e]);
+    m = wave_reduce_max4(m);
+    float4 s = make_float4(0.f, 0.f, 0.f, 0.f);
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) {
+      const float4 v = l1[e];
+      s.x += __expf(v.x - m.x); s.y += __expf(v.y - m.y);
+      s.z += __expf(v.z - m.z); s.w += __expf(v.w - m.w);
+    }
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) {
+      const float4 v = l2[e];
+      s.x += __expf(v.x - m.x); s.y += __expf(v.y - m.y);
+      s.z += __expf(v.z - m.z); s.w += __expf(v.w - m.w);
+    }
+    s = wave_reduce_sum4(s);
+    const float4 inv = make_float4(
+        1.0f / fmaxf(s.x, 1e-38f), 1.0f / fmaxf(s.y, 1e-38f),
+        1.0f / fmaxf(s.z, 1e-38f), 1.0f / fmaxf(s.w, 1e-38f));
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) {
+      const float4 v = l1[e];
+      float4 a;
+      a.x = __expf(v.x - m.x) * inv.x; a.y = __expf(v.y - m.y) * inv.y;
+      a.z = __expf(v.z - m.z) * inv.z; a.w = __expf(v.w - m.w) * inv.w;
+      a1[e] = a;
+      if (drop) {
+        float4 d;
+        d.x = drop_keep(e * 4 + 0, thr, seed) ? a.x * inv_keep : 0.f;
+        d.y = drop_keep(e * 4 + 1, thr, seed) ? a.y * inv_keep : 0.f;
+        d.z = drop_keep(e * 4 + 2, thr, seed) ? a.z * inv_keep : 0.f;
+        d.w = drop_keep(e * 4 + 3, thr, seed) ? a.w * inv_keep : 0.f;
+        da1[e] = d;
+      }
+    }
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) {
+      const float4 v = l2[e];
+      float4 a;
+      a.x = __expf(v.x - m.x) * inv.x; a.y = __expf(v.y - m.y) * inv.y;
+      a.z = __expf(v.z - m.z) * inv.z; a.w = __expf(v.w - m.w) * inv.w;
+      a2[e] = a;
+      if (drop) {
+        const int64_t i = off2 + e * 4;
+        float4 d;
+        d.x = drop_keep(i + 0, thr, seed) ? a.x * inv_keep : 0.f;
+        d.y = drop_keep(i + 1, thr, seed) ? a.y * inv_keep : 0.f;
+        d.z = drop_keep(i + 2, thr, seed) ? a.z * inv_keep : 0.f;
+        d.w = drop_keep(i + 3, thr, seed) ? a.w * inv_keep : 0.f;
+        da2[e] = d;
+      }
+    }
+  }
+}
+
+__global__ void segment_softmax2_h4_bwd_kernel(
+    const int64_t* __restrict__ ip1, const float4* __restrict__ a1,
+    const float4* __restrict__ g1, const int64_t* __restrict__ ip2,
+    const float4* __restrict__ a2, const float4* __restrict__ g2,
+    float4* __restrict__ d1, float4* __restrict__ d2,
+    int n_rows, float keep, uint64_t seed, int64_t off2) {
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+  const bool drop = keep < 1.0f;
+  const uint32_t thr = (uint32_t)(keep * 4294967296.0);
+  const float inv_keep = drop ? 1.0f / keep : 1.0f;
+  for (int r = wave; r < n_rows; r += n_waves) {
+    const int64_t b1 = ip1[r], e1 = ip1[r + 1];
+    const int64_t b2 = ip2[r], e2 = ip2[r + 1];
+    float4 s = make_float4(0.f, 0.f, 0.f, 0.f);
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) {
+      const float4 av = a1[e];
+      float4 gv = g1[e];
+      if (drop) {
+        gv.x = drop_keep(e * 4 + 0, thr, seed) ? gv.x * inv_keep : 0.f;
+        gv.y = drop_keep(e * 4 + 1, thr, seed) ? gv.y * inv_keep : 0.f;
+        gv.z = drop_keep(e * 4 + 2, thr, seed) ? gv.z * inv_keep : 0.f;
+        gv.w = drop_keep(e * 4 + 3, thr, seed) ? gv.w * inv_keep : 0.f;
+      }
+      s.x += av.x * gv.x; s.y += av.y * gv.y;
+      s.z += av.z * gv.z; s.w += av.w * gv.w;
+    }
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) {
+      const float4 av = a2[e];
+      float4 gv = g2[e];
+      if (drop) {
+        const int64_t i = off2 + e * 4;
+        gv.x = drop_keep(i + 0, thr, seed) ? gv.x * inv_keep : 0.f;
+        gv.y = drop_keep(i + 1, thr, seed) ? gv.y * inv_keep : 0.f;
+        gv.z = drop_keep(i + 2, thr, seed) ? gv.z * inv_keep : 0.f;
+        gv.w = drop_keep(i + 3, thr, seed) ? gv.w * inv_keep : 0.f;
+      }
+      s.x += av.x * gv.x; s.y += av.y * gv.y;
+      s.z += av.z * gv.z; s.w += av.w * gv.w;
+    }
+    s = wave_reduce_sum4(s);
+    for (int64_t e = b1 + lane; e < e1; e += WAVE) {
+      const float4 av = a1[e];
+      float4 gv = g1[e];
+      if (drop) {
+        gv.x = drop_keep(e * 4 + 0, thr, seed) ? gv.x * inv_keep : 0.f;
+        gv.y = drop_keep(e * 4 + 1, thr, seed) ? gv.y * inv_keep : 0.f;
+        gv.z = drop_keep(e * 4 + 2, thr, seed) ? gv.z * inv_keep : 0.f;
+        gv.w = drop_keep(e * 4 + 3, thr, seed) ? gv.w * inv_keep : 0.f;
+      }
+      d1[e] = make_float4(av.x * (gv.x - s.x), av.y * (gv.y - s.y),
+                          av.z * (gv.z - s.z), av.w * (gv.w - s.w));
+    }
+    for (int64_t e = b2 + lane; e < e2; e += WAVE) {
+      const float4 av = a2[e];
+      float4 gv = g2[e];
+      if (drop) {
+        const int64_t i = off2 + e * 4;
+        gv.x = drop_keep(i + 0, thr, seed) ? gv.x * inv_keep : 0.f;
+        gv.y = drop_keep(i + 1, thr, seed) ? gv.y * inv_keep : 0.f;
+        gv.z = drop_keep(i + 2, thr, seed) ? gv.z * inv_keep : 0.f;
+        gv.w = drop_keep(i + 3, thr, seed) ? gv.w * inv_keep : 0.f;
+      }
+      d2[e] = make_float4(av.x * (gv.x - s.x), av.y * (gv.y - s.y),
+                          av.z * (gv.z - s.z), av.w * (gv.w - s.w));
+    }
+  }
+}
+
 // Interleaved variant for H a power of two (<= 32): ONE wave per row
 // covers ALL heads — lane l handles (edge = l >> log2H, head = l & (H-1)),
 // so loads of the [E, H] logits are fully COALESCED (the per-(row,head)
@@ -1670,6 +1843,20 @@ std::vector<at::Tensor> segment_softmax2(at::Tensor ip1, at::Tensor l1,
   // into per-row ones — measured SLOWER on Yelp GAT (85 -> 112 ms epoch;
   // short power-law segments are latency-bound and lose the 4x head
   // parallelism). Kept behind BNSGCN_SOFTMAX_ILV=1.
+  if (H == 4) {   // float4-lane form: coalesced, same pass structure
+    hipLaunchKernelGGL(segment_softmax2_h4_kernel,
+                       dim3(spmm_grid(n_rows)), dim3(256), 0, stream,
+                       ip1.data_ptr<int64_t>(),
+                       reinterpret_cast<const float4*>(l1.data_ptr<float>()),
+                       ip2.data_ptr<int64_t>(),
+                       reinterpret_cast<const float4*>(l2.data_ptr<float>()),
+                       reinterpret_cast<float4*>(a1.data_ptr<float>()),
+                       reinterpret_cast<float4*>(a2.data_ptr<float>()),
+                       reinterpret_cast<float4*>(da1.data_ptr<float>()),
+                       reinterpret_cast<float4*>(da2.data_ptr<float>()),
+                       n_rows, (float)keep, (uint64_t)seed, l1.numel());
+    return {a1, a2, da1, da2};
+  }
   static const int ilv_env = [] {
     const char* e = getenv("BNSGCN_SOFTMAX_ILV");
     return e ? atoi(e) : 0;
@@ -1705,6 +1892,20 @@ std::vector<at::Tensor> segment_softmax2_backward(at::Tensor ip1, at::Tensor a1,
   auto d2 = at::empty_like(a2);
   if (n_rows == 0) return {d1, d2};
   auto stream = at::cuda::getCurrentCUDAStream();
+  if (H == 4) {
+    hipLaunchKernelGGL(segment_softmax2_h4_bwd_kernel,
+                       dim3(spmm_grid(n_rows)), dim3(256), 0, stream,
+                       ip1.data_ptr<int64_t>(),
+                       reinterpret_cast<const float4*>(a1.data_ptr<float>()),
+                       reinterpret_cast<const float4*>(g1.data_ptr<float>()),
+                       ip2.data_ptr<int64_t>(),
+                       reinterpret_cast<const float4*>(a2.data_ptr<float>()),
+                       reinterpret_cast<const float4*>(g2.data_ptr<float>()),
+                       reinterpret_cast<float4*>(d1.data_ptr<float>()),
+                       reinterpret_cast<float4*>(d2.data_ptr<float>()),
+                       n_rows, (float)keep, (uint64_t)seed, a1.numel());
+    return {d1, d2};
+  }
   static const int ilv_env = [] {
     const char* e = getenv("BNSGCN_SOFTMAX_ILV");
     return e ? atoi(e) : 0;
