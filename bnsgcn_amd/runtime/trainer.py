@@ -67,7 +67,6 @@ def _load_data_cached(args):
                   name=args.dataset)
     else:
         g = load_data(args.dataset, seed=args.seed, scale=args.data_scale)
-        extra = {}
         if isinstance(g.feat, LazyFeat):
             extra = {"feat": np.zeros((0, g.n_feat), dtype=np.float32),
                      "procedural_seed": g.feat.seed, "procedural_nf": g.n_feat}
