@@ -183,8 +183,8 @@ def main():
     assert torch.isfinite(final_loss), \
         f"non-finite loss after {a.steps} steps — bench result invalid"
 
-    # MAX over ranks
-    t = torch.tensor([elapsed])
+    # MAX over ranks (NCCL needs a device-resident tensor)
+    t = torch.tensor([elapsed], device=device if cuda else "cpu")
     if world > 1:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t[0])

@@ -119,6 +119,8 @@ def exchange_counts(my_counts: torch.Tensor) -> torch.Tensor:
     dev = my_counts.device
     if dist.get_backend() == "gloo" and my_counts.is_cuda:
         my_counts = my_counts.cpu()   # gloo all_gather is CPU-only
+    elif dist.get_backend() == "nccl" and not my_counts.is_cuda:
+        my_counts = my_counts.cuda()  # RCCL is device-only
     gathered = [torch.zeros_like(my_counts) for _ in range(size)]
     dist.all_gather(gathered, my_counts)
     return torch.stack(gathered)[:, rank].to(dev)

@@ -466,7 +466,9 @@ def run(args, rank: int | None = None, world_size: int | None = None) -> dict:
         # rank must fall back together, else dist_evaluate's collectives
         # deadlock against thread-mode ranks
         if world > 1:
-            flag = torch.tensor([have_stores])
+            flag = torch.tensor([have_stores],    # NCCL: device-resident
+                                device=device if str(device).startswith("cuda")
+                                else "cpu")
             dist.all_reduce(flag, op=dist.ReduceOp.MIN)
             have_stores = float(flag[0])
         if have_stores < 1.0:
