@@ -1879,8 +1879,12 @@ std::vector<at::Tensor> segment_softmax2(at::Tensor ip1, at::Tensor l1,
   // short power-law segments are latency-bound and lose the 4x head
   // parallelism). Kept behind BNSGCN_SOFTMAX_ILV=1.
   if (H == 4) {   // float4-lane form: coalesced, same pass structure
+    // short power-law segments are latency-bound: give each wave as few
+    // serial rows as possible (the 16384-block worklist cap left ~11
+    // rows chained per wave)
+    const int g4 = (int)std::min<int64_t>(((int64_t)n_rows + 3) / 4, 65536);
     hipLaunchKernelGGL(segment_softmax2_h4_kernel,
-                       dim3(spmm_grid(n_rows)), dim3(256), 0, stream,
+                       dim3(std::max(g4, 1)), dim3(256), 0, stream,
                        ip1.data_ptr<int64_t>(),
                        reinterpret_cast<const float4*>(l1.data_ptr<float>()),
                        ip2.data_ptr<int64_t>(),
@@ -1928,8 +1932,9 @@ std::vector<at::Tensor> segment_softmax2_backward(at::Tensor ip1, at::Tensor a1,
   if (n_rows == 0) return {d1, d2};
   auto stream = at::cuda::getCurrentCUDAStream();
   if (H == 4) {
+    const int g4 = (int)std::min<int64_t>(((int64_t)n_rows + 3) / 4, 65536);
     hipLaunchKernelGGL(segment_softmax2_h4_bwd_kernel,
-                       dim3(spmm_grid(n_rows)), dim3(256), 0, stream,
+                       dim3(std::max(g4, 1)), dim3(256), 0, stream,
                        ip1.data_ptr<int64_t>(),
                        reinterpret_cast<const float4*>(a1.data_ptr<float>()),
                        reinterpret_cast<const float4*>(g1.data_ptr<float>()),
