@@ -65,7 +65,11 @@ class _SyncBNFunc(Function):
         red = torch.stack((grad.sum(0), (grad * x_hat).sum(0)))
         _maybe_all_reduce(red)
         dbias, dweight = red[0], red[1]
-        dx = (weight / n) / std * (n * grad - dbias - x_hat * dweight)
+        # standard BN input gradient: the batch couples every row through
+        # mean/var, so dx = w/std * (g - mean_rows(g) - x_hat*mean_rows(g·x_hat))
+        g_mean = dbias / n
+        gx_mean = dweight / n
+        dx = weight / std * (grad - g_mean - x_hat * gx_mean)
         return dx, dweight, dbias, None, None, None, None, None, None
 
 
