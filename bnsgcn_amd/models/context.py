@@ -177,16 +177,24 @@ class GraphContext:
             _worklist_of(hbip)
         return (hfip, hfix, hbip, hbix)
 
-    def prefetch_rows_halo(self, st: EpochState, stream) -> None:
+    def prefetch_rows_halo(self, st: EpochState, stream,
+                           gat: bool = False) -> None:
         """Build the restricted-halo CSRs for a FUTURE epoch state on the
         given side stream (called by RankState.prefetch alongside
-        HaloPlan.prefetch)."""
+        HaloPlan.prefetch). gat=True builds the GAT edge-set variant
+        (with the transpose edge permutation) instead."""
         if self.loss_rows is None:
             return
         with torch.cuda.stream(stream):
-            out = self._build_rows_halo(self.loss_rows, st)
+            if gat:
+                out = self._build_gat_rows_halo(st)
+            else:
+                out = self._build_rows_halo(self.loss_rows, st)
             ev = stream.record_event()
-        self._rows_halo = (st.epoch, out, ev)
+        if gat:
+            self._gat_rows_halo = (st.epoch, out, ev)
+        else:
+            self._rows_halo = (st.epoch, out, ev)
 
     # ------------------------------------------------------------- GAT block
     def gat_block(self):
@@ -254,10 +262,27 @@ class GraphContext:
 
     def gat_rows_halo(self, st: EpochState):
         """Row-gathered sampled-halo edge set for the restricted final
-        GAT layer, cached per epoch."""
+        GAT layer, cached per epoch (optionally pre-built on the prefetch
+        stream — see prefetch_rows_halo)."""
         c = getattr(self, "_gat_rows_halo", None)
         if c is not None and c[0] == st.epoch:
+            ev = c[2]
+            if ev is not None:        # built on the prefetch stream
+                cur = torch.cuda.current_stream()
+                cur.wait_event(ev)
+                for t in c[1]:
+                    t.record_stream(cur)
+                    wl = getattr(t, "_bns_worklist", None)
+                    if wl is not None:
+                        for w in wl:
+                            w.record_stream(cur)
+                self._gat_rows_halo = (st.epoch, c[1], None)
             return c[1]
+        out = self._build_gat_rows_halo(st)
+        self._gat_rows_halo = (st.epoch, out, None)
+        return out
+
+    def _build_gat_rows_halo(self, st: EpochState):
         from ..ops.csr_torch import gather_rows_csr
         rows = self.loss_rows
         hip_, hix_ = gather_rows_csr(st.halo_fwd_indptr,
@@ -268,9 +293,7 @@ class GraphContext:
             from ..ops.functional import _worklist_of
             _worklist_of(hip_)
             _worklist_of(hbip)
-        out = (hip_, hix_, hbip, hbix, eperm_t)
-        self._gat_rows_halo = (st.epoch, out)
-        return out
+        return (hip_, hix_, hbip, hbix, eperm_t)
 
     def gat_split_full(self):
         """Static FULL-halo edge set (GAT layer 0 under use_pp)."""

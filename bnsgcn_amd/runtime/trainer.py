@@ -171,7 +171,8 @@ class RankState:
         self.plan.prefetch(epoch)
         nxt = getattr(self.plan, "_next", None)
         if nxt is not None and self.ctx.loss_rows is not None:
-            self.ctx.prefetch_rows_halo(nxt[1], self.plan._prefetch_stream)
+            self.ctx.prefetch_rows_halo(nxt[1], self.plan._prefetch_stream,
+                                        gat=(self.args.model == "gat"))
 
     # ---------------------------------------------------------- precompute
     @torch.no_grad()
