@@ -371,6 +371,57 @@ def test_disk_ingestion_end_to_end_training(tmp_path):
         assert np.isfinite(r["loss_history"]).all()
 
 
+def test_rank_with_zero_train_nodes(tmp_path):
+    """A partition can hold NO labeled nodes on real data splits. The
+    rank must still train (loss 0, zero grads, collectives aligned) —
+    including through the final-layer loss-row restriction, whose
+    restricted CSRs then have zero rows."""
+    import sys
+    sys.path.insert(0, os.path.dirname(__file__))
+    from util_dist import run_dist
+    from bnsgcn_amd.runtime.config import create_parser, graph_name_of
+    from bnsgcn_amd.runtime.trainer import prepare_partitions
+
+    rng = np.random.default_rng(2)
+    n = 80
+    src = rng.integers(0, n, 500)
+    dst = rng.integers(0, n, 500)
+    feat = rng.standard_normal((n, 5)).astype(np.float32)
+    lab = rng.integers(0, 3, n)
+    tm = np.zeros(n, dtype=bool)
+    tm[:40] = True                      # all train nodes in the first half
+    vm = np.zeros(n, dtype=bool)
+    vm[40:60] = True
+    datadir = tmp_path / "d"
+    os.makedirs(datadir)
+    np.savez(datadir / "mini0.npz", src=src, dst=dst, feat=feat, label=lab,
+             train_mask=tm, val_mask=vm, test_mask=~(tm | vm))
+
+    args = create_parser().parse_args([])
+    args.dataset = "mini0"
+    args.data_path = str(datadir)
+    args.n_partitions = 2
+    args.partition_method = "contiguous"   # rank 1 = nodes 40..79: 0 train
+    args.n_hidden = 8
+    args.n_layers = 2
+    args.n_epochs = 4
+    args.model = "graphsage"
+    args.use_pp = True
+    args.sampling_rate = 1.0
+    args.eval = False
+    args.backend = "gloo"
+    args.device = "cpu"
+    args.partition_dir = str(tmp_path / "p")
+    args.graph_name = graph_name_of(args)
+    prepare_partitions(args)
+
+    from test_train_cpu import _train
+    res = run_dist(2, _train, (args,))
+    assert np.isfinite(res[0]["loss_history"]).all()
+    # the empty rank reports loss 0 over max(part_train,1)
+    assert np.isfinite(res[1]["loss_history"]).all()
+
+
 def test_parity_doc_paths_exist():
     """PARITY.md is the judge-facing component map — every repo path it
     cites must exist."""
