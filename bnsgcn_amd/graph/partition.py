@@ -16,9 +16,11 @@ delegates to DGL/METIS). libmetis/DGL are not available in this image, so:
   benchmarks on the planted-locality synthetic graphs;
 * method="bfs": balanced multi-source BFS region growing (single-level).
 
-The partition objective flag (vol/cut) is accepted and recorded in
-meta.json; the multilevel algorithm minimizes edge cut (which on these
-bounded-degree graphs tracks comm volume closely).
+The partition objective flag (vol/cut, reference partition_graph
+objtype) steers the multilevel refinement: "cut" ranks moves by edge-cut
+gain; "vol" ranks by first-order communication-volume gain (distinct
+foreign neighbor partitions per node — exactly the per-layer BNS payload
+unit) with cut as the tiebreak.
 """
 from __future__ import annotations
 
@@ -29,7 +31,7 @@ from .store import Partition, save_partitions
 
 
 def assign_parts(n_nodes: int, n_parts: int, method: str, seed: int = 0,
-                 adj=None) -> np.ndarray:
+                 adj=None, objective: str = "cut") -> np.ndarray:
     if n_parts == 1:
         return np.zeros(n_nodes, dtype=np.int32)
     if method == "random":
@@ -45,7 +47,7 @@ def assign_parts(n_nodes: int, n_parts: int, method: str, seed: int = 0,
         return part
     if method == "metis":
         assert adj is not None, "multilevel partitioning needs the adjacency"
-        return _multilevel_parts(adj, n_parts, seed)
+        return _multilevel_parts(adj, n_parts, seed, objective)
     if method == "bfs":
         assert adj is not None, "bfs partitioning needs the adjacency"
         return _bfs_grow_parts(adj, n_parts, seed)
@@ -270,10 +272,18 @@ def _coarse_partition(indptr, indices, w, node_w, n_parts, seed) -> np.ndarray:
     return part
 
 
-def _refine(indptr, indices, w, node_w, part, n_parts, cap_w, rounds, seed):
+def _refine(indptr, indices, w, node_w, part, n_parts, cap_w, rounds, seed,
+            objective: str = "cut"):
     """Gain-based boundary refinement (parallel FM-lite): move nodes to
     the neighboring partition with the largest connectivity gain, damped
-    (p=0.6) against oscillation, target-capacity enforced in gain order."""
+    (p=0.6) against oscillation, target-capacity enforced in gain order.
+
+    objective="vol" (reference: DGL partition_graph's objtype, helper/
+    utils.py:94): first-order communication-volume gain — a node's vol
+    contribution is its count of DISTINCT foreign neighbor partitions, so
+    moving u to a part its neighbors already occupy removes one (u, part)
+    boundary pair. Ranked lexicographically (vol gain, then edge-cut
+    gain); "cut" ranks by edge-cut weight alone."""
     n = len(indptr) - 1
     if n == 0:
         return part
@@ -293,6 +303,14 @@ def _refine(indptr, indices, w, node_w, part, n_parts, cap_w, rounds, seed):
         conn[rows_id, part] = -1.0
         bestp = conn.argmax(1)
         gain = conn[rows_id, bestp] - cur
+        if objective == "vol":
+            # +1 if the target part already appears among u's neighbors
+            # (the (u, target) boundary pair disappears), -1 if u still
+            # has neighbors in its current part (a (u, old) pair appears)
+            vgain = ((conn[rows_id, bestp] > 0).astype(np.float64)
+                     - (cur > 0))
+            big = float(np.abs(gain).max()) + 1.0
+            gain = vgain * big + gain
         cand = np.flatnonzero((gain > 0) & (rng.random(n) < 0.6))
         if len(cand) == 0:
             break
@@ -309,7 +327,8 @@ def _refine(indptr, indices, w, node_w, part, n_parts, cap_w, rounds, seed):
     return part
 
 
-def _multilevel_parts(adj, n_parts: int, seed: int) -> np.ndarray:
+def _multilevel_parts(adj, n_parts: int, seed: int,
+                      objective: str = "cut") -> np.ndarray:
     """Multilevel k-way partitioning (the role DGL/METIS plays for the
     reference, helper/utils.py:94): capped-LP star coarsening until
     ~24·P clusters, greedy weighted growth at the coarsest level, then
@@ -355,13 +374,13 @@ def _multilevel_parts(adj, n_parts: int, seed: int) -> np.ndarray:
     part = _coarse_partition(indptr, indices, w, node_w, n_parts, seed)
     cap = int(np.ceil(1.05 * total_w / n_parts))
     part = _refine(indptr, indices, w, node_w, part, n_parts, cap,
-                   rounds=4, seed=seed + 1)
+                   rounds=4, seed=seed + 1, objective=objective)
     for li, (ip, ix, ww, nw, lab) in enumerate(reversed(levels)):
         part = part[lab]                      # project to the finer level
         part = _refine(ip, ix, ww, nw, part, n_parts, cap,
                        rounds=1 if len(ix) > 100_000_000
                        else (2 if len(ix) > 20_000_000 else 3),
-                       seed=seed + 2 + li)
+                       seed=seed + 2 + li, objective=objective)
     return part
 
 
@@ -394,7 +413,8 @@ def partition_graph(g: Graph, n_parts: int, method: str = "metis", seed: int = 0
         )
         p0.boundary = [np.zeros(0, dtype=np.int32)]
         return [p0], meta
-    part = assign_parts(n, n_parts, method, seed, adj=g.adj_in)
+    part = assign_parts(n, n_parts, method, seed, adj=g.adj_in,
+                        objective=objective)
 
     # inner-local id of every node within its partition (sorted-global order)
     inner_local = np.zeros(n, dtype=np.int64)

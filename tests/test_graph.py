@@ -431,3 +431,25 @@ def test_parity_doc_paths_exist():
     for m in set(re.findall(r"(?:bnsgcn_amd|tools|scripts|docs)/[\w./]+", text)):
         path = m.rstrip(".")
         assert os.path.exists(os.path.join(root, path)), path
+
+
+def test_partition_objective_vol_reduces_comm_volume():
+    """--partition-obj vol must produce comm volume (total boundary
+    (node, consumer-part) pairs — the BNS per-layer payload unit) no
+    worse than the cut objective, while staying balanced."""
+    from bnsgcn_amd.graph.partition import assign_parts
+    g = load_data("tiny", seed=17)
+    n, P = g.n_nodes, 4
+    s, d = g.adj_in.to_edges()
+
+    def comm_volume(part):
+        key = np.unique(s.astype(np.int64) * P + part[d])
+        owners = part[(key // P).astype(np.int64)]
+        return int((owners != (key % P)).sum())
+
+    pc = assign_parts(n, P, "metis", seed=0, adj=g.adj_in, objective="cut")
+    pv = assign_parts(n, P, "metis", seed=0, adj=g.adj_in, objective="vol")
+    vol_c, vol_v = comm_volume(pc), comm_volume(pv)
+    assert vol_v <= vol_c * 1.05, (vol_v, vol_c)
+    counts = np.bincount(pv, minlength=P)
+    assert counts.min() > 0 and counts.max() <= 1.10 * n / P
