@@ -145,13 +145,14 @@ def test_bench_8rank_reddit_gloo_cuda_oversubscribed():
 @pytest.mark.gpu
 def test_bench_multirank_gloo_cuda_oversubscribed():
     """2 ranks sharing cuda:0 over gloo (host-staged payloads): real HIP
-    kernels + multi-rank halo exchange + bucketed all-reduce on ONE GPU —
-    the closest 1-GPU stand-in for the 8-rank RCCL job (reference gloo
-    oversubscription, main.py:45)."""
+    kernels + multi-rank halo exchange (with the bf16 wire dtype) +
+    bucketed all-reduce on ONE GPU — the closest 1-GPU stand-in for the
+    8-rank RCCL job (reference gloo oversubscription, main.py:45)."""
     if not torch.cuda.is_available():
         pytest.skip("needs a GPU")
     res = _run_bench_subprocess(2, extra=["--sampling-rate", "0.5",
-                                          "--backend", "gloo"],
+                                          "--backend", "gloo",
+                                          "--halo-dtype", "bf16"],
                                 device="cuda:0")
     assert res["n_gpus"] == 2
     assert np.isfinite(res["value"]) and res["value"] > 0
