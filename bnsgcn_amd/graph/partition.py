@@ -450,6 +450,10 @@ def partition_graph(g: Graph, n_parts: int, method: str = "metis", seed: int = 0
         np.cumsum(counts, out=inner_indptr[1:])
         inner_csr = CSR(inner_indptr, inner_local[e_src[im]].astype(np.int32),
                         n_inner)
+        # ascending source ids within each row: the SpMM's shfl-broadcast
+        # batch then walks x row addresses monotonically (prefetch- and
+        # L2-friendlier than generator chunk order)
+        inner_csr.sort_within_rows()
 
         # halo edges grouped by (owner, owner-local id)
         hm = ~im
