@@ -86,8 +86,6 @@ def save_partitions(parts: list[Partition], meta: dict, out_dir: str, graph_name
         meta = dict(meta)
         meta["procedural_feat"] = {"seed": parts[0].feat.seed,
                                    "n_feat": parts[0].feat.shape[1]}
-    with open(os.path.join(d, "meta.json"), "w") as f:
-        json.dump(meta, f, indent=1)
     for p in parts:
         arrs = {k: getattr(p, k) for k in (
             "inner_global_nid", "feat", "label", "train_mask", "val_mask", "test_mask",
@@ -99,6 +97,10 @@ def save_partitions(parts: list[Partition], meta: dict, out_dir: str, graph_name
         for j, b in enumerate(p.boundary):
             arrs[f"boundary_{j}"] = b
         np.savez(os.path.join(d, f"part{p.rank}.npz"), **arrs)
+    # meta.json LAST: its presence marks the store complete (skip_partition
+    # checks it — a crash mid-save must not leave a trusted partial store)
+    with open(os.path.join(d, "meta.json"), "w") as f:
+        json.dump(meta, f, indent=1)
     return d
 
 
