@@ -37,7 +37,8 @@ def create_parser() -> argparse.ArgumentParser:
     p.add_argument("--node-rank", "--node_rank", type=int, default=0)
     p.add_argument("--parts-per-node", "--parts_per_node", type=int, default=10)
     p.add_argument("--partition-method", "--partition_method", type=str,
-                   default="metis", choices=["metis", "random", "bfs"])
+                   default="metis",
+                   choices=["metis", "random", "bfs", "contiguous"])
     p.add_argument("--partition-obj", "--partition_obj", type=str,
                    default="vol", choices=["vol", "cut"])
     p.add_argument("--skip-partition", "--skip_partition", action="store_true")
