@@ -379,3 +379,16 @@ def test_comm_timer_cpu_spans():
     assert 0.015 < tot < 0.5
     t.clear()
     assert t.tot_time() == 0.0
+
+
+def test_graph_name_includes_nonunit_scale():
+    """Partition-store names must encode a non-unit --data-scale so
+    differently-scaled stores never collide (a 0.125-scale store was
+    silently reused by a full-scale bench run before this)."""
+    from bnsgcn_amd.runtime.config import create_parser, graph_name_of
+    args = create_parser().parse_args([])
+    assert graph_name_of(args) == "reddit-2-metis-vol-trans"
+    args.data_scale = 0.125
+    assert graph_name_of(args) == "reddit-2-metis-vol-trans-x0.125"
+    args.graph_name = "explicit"
+    assert graph_name_of(args) == "explicit"

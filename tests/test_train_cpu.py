@@ -149,15 +149,17 @@ def test_multilabel_bce(tmp_path):
         assert lh[-1] < lh[0]
 
 
-def test_bf16_halo_wire_close_to_fp32(tmp_path):
+@pytest.mark.parametrize("model", ["graphsage", "gat"])
+def test_bf16_halo_wire_close_to_fp32(tmp_path, model):
     """--halo-dtype bf16 (wire-only downcast of the halo payloads) must
     track the fp32 trajectory closely — compute and the 1/ratio estimator
-    stay fp32, only the exchanged rows are rounded."""
-    f32 = _run_config(tmp_path / "a", 2, model="graphsage",
-                      sampling_rate=0.5, use_pp=True, n_epochs=10)
-    bf = _run_config(tmp_path / "b", 2, model="graphsage",
-                     sampling_rate=0.5, use_pp=True, n_epochs=10,
-                     halo_dtype="bf16")
+    stay fp32, only the exchanged rows are rounded. GAT covers the
+    raw-feature halo exchange path too."""
+    kw = dict(model=model, sampling_rate=0.5, use_pp=True, n_epochs=10)
+    if model == "gat":
+        kw.update(heads=2, n_hidden=8)
+    f32 = _run_config(tmp_path / "a", 2, **kw)
+    bf = _run_config(tmp_path / "b", 2, halo_dtype="bf16", **kw)
     a = np.array(f32[0]["loss_history"]) + np.array(f32[1]["loss_history"])
     b = np.array(bf[0]["loss_history"]) + np.array(bf[1]["loss_history"])
     np.testing.assert_allclose(b, a, rtol=0.05)
