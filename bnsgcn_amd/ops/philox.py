@@ -8,7 +8,7 @@ SAME sample independently, so the per-epoch ID exchange disappears
 entirely, and sampling is reproducible (fixing reference quirk SURVEY.md
 §2.5.7: numpy was never seeded).
 
-The HIP kernel (ops/hip/bns_kernels.hip: philox_keys_kernel) implements the
+The HIP kernel (ops/hip/kernels.hip: philox_keys_kernel) implements the
 identical function; tests assert bitwise equality between this numpy
 implementation and the device kernel.
 """
