@@ -131,6 +131,78 @@ __global__ __launch_bounds__(256) void spmm_sum_vec4_kernel(
   }
 }
 
+// float2 form for F % 2 == 0 (but % 4 != 0, e.g. the 602-wide Reddit
+// feature matrix: rows are 8-byte but not 16-byte aligned): 4 float2
+// accumulators cover 512 floats per edge walk — the scalar form needed
+// 3 walks at 4-byte loads for F=602 (57 ms for the one-time precompute
+// SpMM, profiles/topk_reddit_final_r02.txt).
+template <bool ACC>
+__global__ __launch_bounds__(256) void spmm_sum_vec2_kernel(
+    const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
+    const int64_t* __restrict__ wend, const int32_t* __restrict__ wave_start,
+    const int32_t* __restrict__ indices, const float* __restrict__ x,
+    const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
+    float* __restrict__ out, int f2) {
+  const int bb = xcd_remap_block(blockIdx.x, gridDim.x);
+  const int w = bb * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int it_beg = wave_start[w], it_end = wave_start[w + 1];
+  const float2* __restrict__ x2 = reinterpret_cast<const float2*>(x);
+  float2* __restrict__ out2 = reinterpret_cast<float2*>(out);
+  for (int it = it_beg; it < it_end; ++it) {
+    int row = wrow[it];
+    const bool atomic = row < 0;
+    if (atomic) row = ~row;
+    const int64_t beg = wbeg[it], end = wend[it];
+    const float ds = dst_scale ? dst_scale[row] : 1.0f;
+    for (int f0 = 0; f0 < f2; f0 += 4 * WAVE) {
+      float2 acc[4] = {{0.f, 0.f}, {0.f, 0.f}, {0.f, 0.f}, {0.f, 0.f}};
+      for (int64_t e0 = beg; e0 < end; e0 += WAVE) {
+        const int nv = (int)((end - e0 < WAVE) ? (end - e0) : WAVE);
+        int cid = 0;
+        float ssc = 1.0f;
+        if (lane < nv) {
+          cid = indices[e0 + lane];
+          if (src_scale) ssc = src_scale[cid];
+        }
+#pragma unroll 4
+        for (int k = 0; k < nv; ++k) {
+          const int c = __shfl(cid, k, WAVE);
+          const float ss = src_scale ? __shfl(ssc, k, WAVE) : 1.0f;
+          const int64_t base = (int64_t)c * f2;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const int f = f0 + j * WAVE + lane;
+            if (f < f2) {
+              const float2 v = x2[base + f];
+              acc[j].x += ss * v.x;
+              acc[j].y += ss * v.y;
+            }
+          }
+        }
+      }
+      const int64_t ob = (int64_t)row * f2;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int f = f0 + j * WAVE + lane;
+        if (f >= f2) break;
+        if (atomic) {
+          float* p = reinterpret_cast<float*>(&out2[ob + f]);
+          atomicAdd(p + 0, ds * acc[j].x);
+          atomicAdd(p + 1, ds * acc[j].y);
+        } else if (ACC) {
+          float2 pv = out2[ob + f];
+          pv.x += ds * acc[j].x;
+          pv.y += ds * acc[j].y;
+          out2[ob + f] = pv;
+        } else {
+          out2[ob + f] = make_float2(ds * acc[j].x, ds * acc[j].y);
+        }
+      }
+    }
+  }
+}
+
 template <bool ACC>
 __global__ __launch_bounds__(256) void spmm_sum_scalar_kernel(
     const int32_t* __restrict__ wrow, const int64_t* __restrict__ wbeg,
@@ -1635,6 +1707,14 @@ at::Tensor spmm_sum(at::Tensor wrow, at::Tensor wbeg, at::Tensor wend,
                        indices.data_ptr<int32_t>(), x.data_ptr<float>(),
                        opt_ptr(src_scale), opt_ptr(dst_scale),
                        out.data_ptr<float>(), F / 4, strided);
+  } else if (F % 2 == 0) {
+    auto kfn = acc ? spmm_sum_vec2_kernel<true> : spmm_sum_vec2_kernel<false>;
+    hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,
+                       wrow.data_ptr<int32_t>(), wbeg.data_ptr<int64_t>(),
+                       wend.data_ptr<int64_t>(), wave_start.data_ptr<int32_t>(),
+                       indices.data_ptr<int32_t>(), x.data_ptr<float>(),
+                       opt_ptr(src_scale), opt_ptr(dst_scale),
+                       out.data_ptr<float>(), F / 2);
   } else {
     auto kfn = acc ? spmm_sum_scalar_kernel<true> : spmm_sum_scalar_kernel<false>;
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(256), 0, stream,

@@ -31,7 +31,7 @@ def rand_csr(n_rows, n_cols, e, seed=0):
 
 
 @needs_gpu
-@pytest.mark.parametrize("F", [256, 602, 17, 512, 128, 64])
+@pytest.mark.parametrize("F", [256, 602, 17, 512, 128, 64, 30])
 def test_spmm_matches_reference(F):
     from bnsgcn_amd.ops.functional import spmm_sum_raw
     torch.manual_seed(0)
