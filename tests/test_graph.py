@@ -453,3 +453,44 @@ def test_partition_objective_vol_reduces_comm_volume():
     assert vol_v <= vol_c * 1.05, (vol_v, vol_c)
     counts = np.bincount(pv, minlength=P)
     assert counts.min() > 0 and counts.max() <= 1.10 * n / P
+
+
+def test_partitioner_internal_helpers():
+    """Property checks for the multilevel building blocks: _row_argmax
+    (per-row argmax over CSR values, empty rows, uniform-weight shortcut)
+    and _grouped_cumsum (restarting inclusive cumsum)."""
+    from bnsgcn_amd.graph.partition import _row_argmax, _grouped_cumsum
+    rng = np.random.default_rng(0)
+    # random CSR with some empty rows
+    lens = rng.integers(0, 5, 50)
+    indptr = np.zeros(51, dtype=np.int64)
+    indptr[1:] = np.cumsum(lens)
+    vals = rng.integers(1, 100, int(lens.sum())).astype(np.int64)
+    out = _row_argmax(indptr, vals)
+    for r in range(50):
+        b, e = indptr[r], indptr[r + 1]
+        if b == e:
+            assert out[r] == -1
+        else:
+            assert b <= out[r] < e
+            assert vals[out[r]] == vals[b:e].max()
+            # first occurrence of the max wins (deterministic)
+            assert out[r] == b + int(np.argmax(vals[b:e]))
+    # uniform-weight shortcut: first column per row
+    ones = np.ones_like(vals)
+    out1 = _row_argmax(indptr, ones)
+    nz = np.flatnonzero(lens > 0)
+    np.testing.assert_array_equal(out1[nz], indptr[:-1][nz])
+
+    # grouped cumsum vs a reference loop
+    groups = np.sort(rng.integers(0, 8, 200))
+    v = rng.integers(0, 10, 200).astype(np.int64)
+    got = _grouped_cumsum(groups, v)
+    run = {}
+    want = np.empty_like(v)
+    for i, (g, x) in enumerate(zip(groups, v)):
+        run[g] = run.get(g, 0) + x
+        want[i] = run[g]
+    np.testing.assert_array_equal(got, want)
+    assert _grouped_cumsum(np.array([], dtype=np.int64),
+                           np.array([], dtype=np.int64)).size == 0
