@@ -494,3 +494,16 @@ def test_partitioner_internal_helpers():
     np.testing.assert_array_equal(got, want)
     assert _grouped_cumsum(np.array([], dtype=np.int64),
                            np.array([], dtype=np.int64)).size == 0
+
+
+def test_partition_quality_tool_smoke():
+    import subprocess, sys
+    r = subprocess.run(
+        [sys.executable,
+         os.path.join(os.path.dirname(__file__), "..", "tools",
+                      "partition_quality.py"),
+         "--dataset", "tiny", "--data-scale", "1.0", "--n-partitions", "3",
+         "--methods", "metis", "random"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "metis" in r.stdout and "comm_volume=" in r.stdout
