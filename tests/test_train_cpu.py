@@ -459,3 +459,28 @@ def test_run_with_dist_eval_inductive(tmp_path):
         assert abs(multi[0]["test_acc"] - want_test) < 1e-5
     finally:
         os.chdir(cwd)
+
+
+def test_main_mpi_branch_raises_without_mpirun(tmp_path):
+    """--backend mpi execs mpirun (reference main.py:51-62); without
+    mpirun on PATH the launcher must fail fast with a clear message."""
+    import shutil
+    import subprocess
+    import sys
+    if shutil.which("mpirun"):
+        pytest.skip("mpirun present")
+    r = subprocess.run(
+        [sys.executable,
+         os.path.join(os.path.dirname(__file__), "..", "main.py"),
+         "--dataset", "tiny", "--n-partitions", "2", "--backend", "mpi",
+         "--n-epochs", "1", "--no-eval",
+         "--partition-dir", str(tmp_path / "p")],
+        capture_output=True, text=True, timeout=180, cwd=str(tmp_path))
+    assert r.returncode != 0
+    assert "mpirun" in r.stderr
+
+
+def test_ingest_missing_file_message(tmp_path):
+    from bnsgcn_amd.graph.ingest import load_disk_data
+    with pytest.raises(FileNotFoundError, match="layout"):
+        load_disk_data("nope", str(tmp_path))
